@@ -1,0 +1,99 @@
+"""GFKB HTTP service fronting the GfkbEngine.
+
+Endpoint parity with /root/reference/services/gfkb/app.py:74-198:
+GET /failures, POST /failures/match, POST /failures/upsert,
+GET /patterns, POST /patterns/upsert. The engine behind them is the
+HBM-resident embedding store + fused cosine-topk kernel (gfkb/engine.py)
+instead of a per-request TF-IDF refit.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Any, Dict, List, Optional
+
+from fastapi import FastAPI
+from pydantic import BaseModel, Field
+
+from kakveda_amd.core.schemas import (
+    FailureMatchRequest,
+    FailureMatchResponse,
+    Severity,
+)
+from kakveda_amd.gfkb.engine import GfkbEngine
+
+
+class UpsertFailureRequest(BaseModel):
+    failure_type: str
+    root_cause: Optional[str] = None
+    context_signature: Dict[str, Any] = Field(default_factory=dict)
+    impact_severity: Severity = Severity.medium
+    resolution: Optional[str] = None
+    signature_text: str
+    app_id: str
+
+
+class UpsertPatternRequest(BaseModel):
+    name: str
+    failure_ids: List[str]
+    affected_apps: List[str]
+    description: Optional[str] = None
+
+
+def create_app(
+    data_dir: Optional[str] = None,
+    device: Optional[str] = None,
+    engine: Optional[GfkbEngine] = None,
+) -> FastAPI:
+    app = FastAPI(title="Kakveda-AMD GFKB")
+    if engine is None:
+        import torch
+
+        dev = device or ("cuda" if torch.cuda.is_available() else "cpu")
+        engine = GfkbEngine(
+            data_dir=data_dir or os.environ.get("DATA_DIR", "/app/data"), device=dev
+        )
+    app.state.engine = engine
+
+    @app.get("/failures")
+    async def list_failures():
+        return {"failures": engine.list_failures()}
+
+    @app.post("/failures/match", response_model=FailureMatchResponse)
+    async def match(req: FailureMatchRequest):
+        return FailureMatchResponse(
+            matches=engine.match(req.signature_text, failure_type=req.failure_type)
+        )
+
+    @app.post("/failures/upsert")
+    async def upsert(req: UpsertFailureRequest):
+        rec, created = engine.upsert_failure(
+            failure_type=req.failure_type,
+            signature_text=req.signature_text,
+            context_signature=req.context_signature,
+            impact_severity=req.impact_severity.value,
+            root_cause=req.root_cause,
+            resolution=req.resolution,
+            app_id=req.app_id,
+        )
+        return {"ok": True, "created": created, "failure": rec}
+
+    @app.get("/patterns")
+    async def list_patterns():
+        return {"patterns": engine.list_patterns()}
+
+    @app.post("/patterns/upsert")
+    async def upsert_pattern(req: UpsertPatternRequest):
+        rec, created = engine.upsert_pattern(
+            name=req.name,
+            failure_ids=req.failure_ids,
+            affected_apps=req.affected_apps,
+            description=req.description,
+        )
+        return {"ok": True, "created": created, "pattern": rec}
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"ok": True, "failures": len(engine.failures), "rows": engine.store.count}
+
+    return app
