@@ -1,0 +1,163 @@
+"""GPU numerics tests: HIP kernels vs plain-PyTorch fp32 references."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+def _rand_unit(n, d, seed=0, device=None):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    x = torch.randn(n, d, generator=g, dtype=torch.float32)
+    x = x / x.norm(dim=-1, keepdim=True)
+    return x.to(device=device or _dev(), dtype=torch.bfloat16)
+
+
+def test_extension_loaded():
+    from kakveda_amd import ops
+
+    assert ops.hip_available(), "HIP extension must be built on GPU boxes"
+
+
+@pytest.mark.parametrize(
+    "B,N,k",
+    [
+        (1, 100, 5),
+        (100, 1000, 5),
+        (128, 4096, 8),
+        (300, 70000, 5),
+        (1024, 300000, 8),
+    ],
+)
+def test_cosine_topk_vs_torch(B, N, k):
+    from kakveda_amd import ops
+
+    D = 768
+    q = _rand_unit(B, D, seed=1)
+    c = _rand_unit(N, D, seed=2)
+
+    scores, idx = ops.cosine_topk(q, c, k)
+    torch.cuda.synchronize()
+
+    # fp32 reference on the SAME bf16-rounded inputs
+    sims = q.float() @ c.float().t()
+    ref_scores, ref_idx = torch.topk(sims, k, dim=1)
+
+    assert scores.shape == (B, k) and idx.shape == (B, k)
+    # scores sorted descending
+    assert (scores[:, :-1] >= scores[:, 1:] - 1e-6).all()
+    # kernel scores match reference top-k scores (bf16 accumulation tolerance)
+    assert torch.allclose(scores, ref_scores, atol=2e-2, rtol=1e-2), (
+        (scores - ref_scores).abs().max().item()
+    )
+    # each claimed index really achieves its claimed score
+    gathered = sims.gather(1, idx.clamp_min(0))
+    assert torch.allclose(gathered, scores, atol=1e-4), (
+        (gathered - scores).abs().max().item()
+    )
+
+
+def test_cosine_topk_valid_n():
+    from kakveda_amd import ops
+
+    q = _rand_unit(16, 768, seed=3)
+    c = _rand_unit(1000, 768, seed=4)
+    # plant a huge match OUTSIDE the valid prefix: must not be returned
+    c[900] = q[0].clone()
+    scores, idx = ops.cosine_topk(q, c, 5, valid_n=800)
+    torch.cuda.synchronize()
+    assert (idx < 800).all()
+    sims = q.float() @ c[:800].float().t()
+    ref_scores, _ = torch.topk(sims, 5, dim=1)
+    assert torch.allclose(scores, ref_scores, atol=2e-2, rtol=1e-2)
+
+
+def test_cosine_topk_exact_match_first():
+    from kakveda_amd import ops
+
+    c = _rand_unit(5000, 768, seed=5)
+    q = c[1234:1235].clone()
+    scores, idx = ops.cosine_topk(q, c, 5)
+    torch.cuda.synchronize()
+    assert idx[0, 0].item() == 1234
+    assert scores[0, 0].item() > 0.99
+
+
+def test_l2normalize():
+    from kakveda_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(9)
+    x = (torch.randn(500, 768, generator=g) * 3.0).to(_dev(), torch.bfloat16)
+    ref = x.float() / x.float().norm(dim=-1, keepdim=True).clamp_min(1e-12)
+    ops.l2normalize_(x)
+    torch.cuda.synchronize()
+    assert torch.allclose(x.float(), ref, atol=1e-2), (x.float() - ref).abs().max()
+    norms = x.float().norm(dim=-1)
+    assert torch.allclose(norms, torch.ones_like(norms), atol=2e-2)
+
+
+def test_l2normalize_partial_rows():
+    from kakveda_amd import ops
+
+    x = torch.ones(10, 768, device=_dev(), dtype=torch.bfloat16) * 2.0
+    ops.l2normalize_(x, start_row=4, end_row=8)
+    torch.cuda.synchronize()
+    assert x[0, 0].item() == 2.0 and x[9, 0].item() == 2.0
+    assert abs(x[5].float().norm().item() - 1.0) < 2e-2
+
+
+def test_embedding_bag_vs_torch():
+    from kakveda_amd import ops
+
+    g = torch.Generator(device="cpu").manual_seed(11)
+    V, D, B, L = 5000, 768, 64, 48
+    table = torch.randn(V, D, generator=g).to(_dev(), torch.bfloat16)
+    idx = torch.randint(0, V, (B, L), generator=g, dtype=torch.int32).to(_dev())
+    w = torch.randn(B, L, generator=g, dtype=torch.float32).to(_dev())
+    w[:, 40:] = 0.0  # padding
+
+    out = ops.embedding_bag(table, idx, w)
+    torch.cuda.synchronize()
+    flat = table.float()[idx.reshape(-1).long()].reshape(B, L, D)
+    ref = (flat * w.unsqueeze(-1)).sum(dim=1)
+    assert torch.allclose(out, ref, atol=5e-2, rtol=1e-2), (out - ref).abs().max()
+
+
+def test_encoder_gpu_matches_cpu():
+    from kakveda_amd.encoder import TraceEncoder
+
+    texts = [
+        "intent_tags:intent:citations_required | prompt_hint:please cite sources | tools: | env_keys:a",
+        "intent_tags: | prompt_hint:what is the weather | tools: | env_keys:b",
+    ]
+    cpu = TraceEncoder(dim=768, hash_dim=8192, seed=3, device="cpu")
+    gpu = TraceEncoder(dim=768, hash_dim=8192, seed=3, device="cuda")
+    e_cpu = cpu.encode_texts(texts)
+    e_gpu = gpu.encode_texts(texts).cpu()
+    torch.cuda.synchronize()
+    # bf16 table/GEMM on GPU vs fp32 CPU: cosine of the two encodings ~ 1
+    cos = (e_cpu * e_gpu).sum(dim=-1)
+    assert (cos > 0.99).all(), cos
+
+
+def test_gfkb_engine_on_gpu(tmp_path):
+    from kakveda_amd.gfkb.engine import GfkbEngine
+
+    eng = GfkbEngine(data_dir=str(tmp_path), device="cuda", dim=768, hash_dim=8192)
+    sig = (
+        "intent_tags:intent:citations_required | prompt_hint:explain with sources"
+        " | tools: | env_keys:x"
+    )
+    eng.upsert_failure("HALLUCINATION_CITATION", sig, {}, app_id="a")
+    for i in range(50):
+        eng.upsert_failure(
+            "T", f"intent_tags: | prompt_hint:filler {i} | tools: | env_keys:", {},
+            app_id="a",
+        )
+    matches = eng.match(sig)
+    assert matches and matches[0].failure_id == "F-0001"
+    assert matches[0].score > 0.98
