@@ -81,7 +81,7 @@ DEVINL bf16x8 read_frag(const char* lds_tile, int row, int slot) {
 //   grid.x = nchunks, grid.y = ceil(B/128), block = 256 threads.
 //   partial_score/partial_idx: [B][nchunks][KMAX]
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(THREADS) void cosine_topk_partial(
+__global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
     float* __restrict__ partial_score, int* __restrict__ partial_idx,
     int B, int N, int D, int chunk_tiles, int nchunks) {
@@ -104,10 +104,31 @@ __global__ __launch_bounds__(THREADS) void cosine_topk_partial(
   const int wr = wid >> 1;        // wave's row quadrant (0/1)
   const int wc = wid & 1;         // wave's col quadrant (0/1)
 
-  const int row0 = blockIdx.y * BM;
+  // XCD-aware remap (guide T1): the dispatcher places dispatch-index b on
+  // XCD b%8. Give each XCD a contiguous span of corpus chunks and iterate
+  // row tiles innermost, so the ~64 co-resident blocks of one XCD work on
+  // the same 1-2 chunks and the corpus tile stream stays in that XCD's L2
+  // instead of being re-fetched from HBM once per query row tile.
+  // Requires gridDim.x % 8 == 0 (the host pads nchunks; padded chunks have
+  // tiles_here <= 0 and just write -inf partials).
+  const int nrt = gridDim.y;
+  int chunk_id, row_tile;
+  if ((gridDim.x & 7) == 0 && gridDim.x * nrt >= 512) {
+    const int bid = blockIdx.x + gridDim.x * blockIdx.y;
+    const int xcd = bid & 7;
+    const int slot = bid >> 3;
+    const int cpx = gridDim.x >> 3;  // chunks per XCD
+    chunk_id = xcd * cpx + slot / nrt;
+    row_tile = slot % nrt;
+  } else {
+    chunk_id = blockIdx.x;
+    row_tile = blockIdx.y;
+  }
+
+  const int row0 = row_tile * BM;
   const long qrow_bytes = (long)D * 2;
   const int ntiles_total = (N + BN - 1) / BN;
-  const int tile0 = blockIdx.x * chunk_tiles;
+  const int tile0 = chunk_id * chunk_tiles;
   const int tiles_here = min(chunk_tiles, ntiles_total - tile0);
   const int nkt = D / BK;
 
@@ -165,62 +186,77 @@ __global__ __launch_bounds__(THREADS) void cosine_topk_partial(
     }
 
     // ---- top-k epilogue -------------------------------------------------
-    // Lane layout of acc (C/D map): col = n*16 + cl, row = m*16 + g*4 + reg
-    // within the wave's 64x64 quadrant. Waves with the same wr share rows,
-    // so the two col-quadrants insert in separate phases.
-    volatile float* vmin = tk_min;
-    volatile float* vscore = tk_score;
-    volatile int* vidx = tk_idx;
-    volatile int* vpos = tk_minpos;
-
+    // Stash the tile's scores into the (now idle) staging LDS, then let
+    // each wave own 32 rows for the insertion pass. This keeps acc out of
+    // the insertion code, which otherwise explodes register pressure
+    // (measured: 147 -> 256 VGPR + 1 KB/lane scratch when the unrolled
+    // insert reads acc directly).
+#ifdef KAKVEDA_NO_EPILOGUE
+    // ablation build: keep acc live, skip the top-k insert (guide rule 17)
 #pragma unroll
-    for (int ph = 0; ph < 2; ++ph) {
-      if (wc == ph) {
+    for (int m = 0; m < 4; ++m)
 #pragma unroll
-        for (int m = 0; m < 4; ++m) {
+      for (int n = 0; n < 4; ++n)
+        asm volatile("" ::"v"(acc[m][n]));
+    if (false)
+#endif
+    {
+      float* stile = (float*)smem0;  // [128][128] f32 over the staging bufs
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
 #pragma unroll
           for (int reg = 0; reg < 4; ++reg) {
             const int row = wr * 64 + m * 16 + g * 4 + reg;
-            float v[4];
-            int gc[4];
-            bool any = false;
-            const float rmin0 = vmin[row];
+            const int col = wc * 64 + n * 16 + cl;
+            stile[row * 128 + col] = acc[m][n][reg];
+          }
+      __syncthreads();
+
+      volatile float* vmin = tk_min;
+      volatile float* vscore = tk_score;
+      volatile int* vidx = tk_idx;
+      volatile int* vpos = tk_minpos;
+
+      for (int rr = 0; rr < 32; ++rr) {
+        const int row = wid * 32 + rr;
+        const float rmin0 = vmin[row];
+        const float v0 = stile[row * 128 + lane];
+        const float v1 = stile[row * 128 + 64 + lane];
+        const int gc0 = col0 + lane;
+        const int gc1 = col0 + 64 + lane;
+        const bool a0 = (gc0 < N) & (v0 > rmin0);
+        const bool a1 = (gc1 < N) & (v1 > rmin0);
+        unsigned long long ball = __ballot(a0 | a1);
+        while (ball) {
+          const int leader = __ffsll((unsigned long long)ball) - 1;
+          if (lane == leader) {
 #pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              gc[n] = col0 + wc * 64 + n * 16 + cl;
-              v[n] = (gc[n] < N) ? acc[m][n][reg] : NEG_INF;
-              any |= v[n] > rmin0;
-            }
-            unsigned long long ball =
-                __ballot(any) & (0xFFFFull << (g * 16));
-            while (ball) {
-              const int leader = __ffsll((unsigned long long)ball) - 1;
-              if (lane == leader) {
+            for (int h = 0; h < 2; ++h) {
+              const float v = h ? v1 : v0;
+              const int gc = h ? gc1 : gc0;
+              const bool a = h ? a1 : a0;
+              if (a && v > vmin[row]) {
+                const int p = vpos[row];
+                vscore[row * KMAX + p] = v;
+                vidx[row * KMAX + p] = gc;
+                float mn = vscore[row * KMAX];
+                int mp = 0;
 #pragma unroll
-                for (int n = 0; n < 4; ++n) {
-                  if (v[n] > vmin[row]) {
-                    const int p = vpos[row];
-                    vscore[row * KMAX + p] = v[n];
-                    vidx[row * KMAX + p] = gc[n];
-                    // rescan for the new min
-                    float mn = vscore[row * KMAX];
-                    int mp = 0;
-#pragma unroll
-                    for (int q = 1; q < KMAX; ++q) {
-                      const float s = vscore[row * KMAX + q];
-                      if (s < mn) { mn = s; mp = q; }
-                    }
-                    vmin[row] = mn;
-                    vpos[row] = mp;
-                  }
+                for (int q = 1; q < KMAX; ++q) {
+                  const float s = vscore[row * KMAX + q];
+                  if (s < mn) { mn = s; mp = q; }
                 }
+                vmin[row] = mn;
+                vpos[row] = mp;
               }
-              ball &= ball - 1;
             }
           }
+          ball &= ball - 1;
         }
       }
-      __syncthreads();
+      __syncthreads();  // lists settled before next tile reuses the LDS
     }
   }
 
@@ -228,7 +264,7 @@ __global__ __launch_bounds__(THREADS) void cosine_topk_partial(
   if (tid < BM) {
     const int grow = row0 + tid;
     if (grow < B) {
-      const size_t base = ((size_t)grow * nchunks + blockIdx.x) * KMAX;
+      const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;
 #pragma unroll
       for (int q = 0; q < KMAX; ++q) {
         partial_score[base + q] = tk_score[tid * KMAX + q];
@@ -413,7 +449,9 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // size chunks so the grid comfortably oversubscribes 256 CUs
   long want = ((long)ntiles * row_tiles + 2047) / 2048;
   const int chunk_tiles = (int)std::max(4L, std::min(want, 128L));
-  const int nchunks = (ntiles + chunk_tiles - 1) / chunk_tiles;
+  // pad the chunk count to a multiple of 8 so the in-kernel XCD remap is
+  // bijective; padded chunks have no tiles and emit -inf partials.
+  const int nchunks = ((ntiles + chunk_tiles - 1) / chunk_tiles + 7) & ~7;
 
   auto opts_f = torch::TensorOptions().dtype(torch::kFloat32).device(queries.device());
   auto opts_i = torch::TensorOptions().dtype(torch::kInt32).device(queries.device());
