@@ -1,0 +1,126 @@
+// Standalone perf probe for the cosine-topk kernel (no torch, no python).
+// Build:  hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/knn_probe.hip -o gpurun_out/knn_probe
+//         (add -DKAKVEDA_NO_EPILOGUE for the GEMM-only ablation)
+// Run:    ./knn_probe [B] [N] [iters]
+// Prints ms/iter and effective TFLOP/s of the score GEMM.
+
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+#include <algorithm>
+
+#include "../kakveda_amd/ops/hip/kernels_impl.h"
+
+using namespace kakveda;
+
+#define HIP_CHECK(x)                                                    \
+  do {                                                                  \
+    hipError_t e = (x);                                                 \
+    if (e != hipSuccess) {                                              \
+      fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(e),  \
+              __FILE__, __LINE__);                                      \
+      exit(1);                                                          \
+    }                                                                   \
+  } while (0)
+
+__global__ void fill_rand(bf16_t* p, size_t n, unsigned seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned h = (unsigned)(i * 2654435761u) ^ seed;
+    h ^= h >> 13; h *= 0x5bd1e995u; h ^= h >> 15;
+    const float f = ((float)(h & 0xFFFF) / 65536.0f - 0.5f) * 0.07217f; // ~unit rows
+    p[i] = (bf16_t)f;
+  }
+}
+
+int main(int argc, char** argv) {
+  const int B = argc > 1 ? atoi(argv[1]) : 4096;
+  const long N = argc > 2 ? atol(argv[2]) : 2000000;
+  const int iters = argc > 3 ? atoi(argv[3]) : 10;
+  const int D = 768, k = 5;
+
+  bf16_t *Q, *C;
+  HIP_CHECK(hipMalloc(&Q, (size_t)B * D * 2));
+  HIP_CHECK(hipMalloc(&C, (size_t)N * D * 2));
+  hipLaunchKernelGGL(fill_rand, dim3(2048), dim3(256), 0, 0, Q, (size_t)B * D, 1u);
+  hipLaunchKernelGGL(fill_rand, dim3(2048), dim3(256), 0, 0, C, (size_t)N * D, 2u);
+
+  const int row_tiles = (B + BM - 1) / BM;
+  const int ntiles = (int)((N + BN - 1) / BN);
+  long want = ((long)ntiles * row_tiles + 2047) / 2048;
+  const int chunk_tiles = (int)std::max(4L, std::min(want, 128L));
+  const int nchunks = ((ntiles + chunk_tiles - 1) / chunk_tiles + 7) & ~7;
+
+  float* pscore;
+  int* pidx;
+  float* oscore;
+  long* oidx;
+  HIP_CHECK(hipMalloc(&pscore, (size_t)B * nchunks * KMAX * 4));
+  HIP_CHECK(hipMalloc(&pidx, (size_t)B * nchunks * KMAX * 4));
+  HIP_CHECK(hipMalloc(&oscore, (size_t)B * k * 4));
+  HIP_CHECK(hipMalloc(&oidx, (size_t)B * k * 8));
+  unsigned* rowthr;
+  HIP_CHECK(hipMalloc(&rowthr, (size_t)B * 4));
+
+  dim3 grid(nchunks, row_tiles);
+  auto run_mode = [&](int mode) {
+    hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0, 0,
+                       rowthr, B);
+    if (mode == 0)
+      hipLaunchKernelGGL((cosine_topk_partial_t<0>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 1)
+      hipLaunchKernelGGL((cosine_topk_partial_t<1>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else
+      hipLaunchKernelGGL((cosine_topk_partial_t<2>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+  };
+
+  for (int m = 0; m < 3; ++m) { run_mode(m); }
+  HIP_CHECK(hipDeviceSynchronize());
+
+  const char* names[3] = {"full", "gemm-only", "precheck+stash"};
+  std::vector<std::vector<float>> ms(3);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  for (int it = 0; it < iters; ++it) {
+    for (int m = 0; m < 3; ++m) {
+      HIP_CHECK(hipEventRecord(t0));
+      run_mode(m);
+      HIP_CHECK(hipEventRecord(t1));
+      HIP_CHECK(hipEventSynchronize(t1));
+      float x;
+      HIP_CHECK(hipEventElapsedTime(&x, t0, t1));
+      ms[m].push_back(x);
+    }
+  }
+  // stats pass
+  unsigned long long* dstats;
+  HIP_CHECK(hipMalloc(&dstats, 4 * 8));
+  HIP_CHECK(hipMemset(dstats, 0, 4 * 8));
+  hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0, 0,
+                     rowthr, B);
+  hipLaunchKernelGGL((cosine_topk_partial_t<3>), grid, dim3(THREADS), 0, 0, Q,
+                     C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks,
+                     rowthr, dstats);
+  HIP_CHECK(hipDeviceSynchronize());
+  unsigned long long hstats[4];
+  HIP_CHECK(hipMemcpy(hstats, dstats, 32, hipMemcpyDeviceToHost));
+  const double tiles = (double)row_tiles * ntiles;
+  printf("stats: stamped=%llu (%.2f/tile) inserts=%llu (%.1f/row/chunk)\n",
+         hstats[0], hstats[0] / tiles, hstats[1],
+         hstats[1] / ((double)B * nchunks));
+  for (int m = 0; m < 3; ++m) {
+    std::sort(ms[m].begin(), ms[m].end());
+    const float med = ms[m][ms[m].size() / 2];
+    const float mn = ms[m][0];
+    const double tf = 2.0 * B * (double)N * D / (med * 1e-3) / 1e12;
+    printf("%-15s B=%d N=%ld med=%.3f ms min=%.3f  %.1f TF\n", names[m], B, N,
+           med, mn, tf);
+  }
+  return 0;
+}
