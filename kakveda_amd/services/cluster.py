@@ -60,6 +60,7 @@ class LocalCluster:
             gfkb_url=u["gfkb"],
             self_url=u["pattern_detector"],
             transport=self.tx,
+            engine=self.engine,
         )
         self.health_scoring = hs_mod.create_app(
             event_bus_url=u["event_bus"],

@@ -26,6 +26,7 @@ def create_app(
     gfkb_url: Optional[str] = None,
     self_url: Optional[str] = None,
     transport: Optional[Transport] = None,
+    engine=None,
 ) -> FastAPI:
     app = FastAPI(title="Kakveda-AMD Pattern Detector")
     bus = event_bus_url or os.environ.get("EVENT_BUS_URL", "http://event-bus:8100")
@@ -77,6 +78,24 @@ def create_app(
             },
         )
         return {"ok": True, "pattern": True}
+
+    @app.post("/cluster/run")
+    async def cluster_run(body: Optional[dict] = None):
+        """Generalised pattern mining: streaming k-means over the GFKB
+        fingerprint embeddings (GPU kernels when the engine is on GPU;
+        RCCL all-reduce of centroid partials when sharded)."""
+        if engine is None:
+            return {"ok": False, "error": "no local engine attached"}
+        from kakveda_amd.patterns.miner import PatternMiner
+
+        body = body or {}
+        miner = PatternMiner(
+            engine,
+            n_clusters=int(body.get("n_clusters", 16)),
+            min_apps=int(body.get("min_apps", 2)),
+        )
+        patterns = miner.mine(iters=int(body.get("iters", 8)))
+        return {"ok": True, "patterns": patterns}
 
     @app.get("/healthz")
     async def healthz():

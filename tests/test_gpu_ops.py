@@ -161,3 +161,24 @@ def test_gfkb_engine_on_gpu(tmp_path):
     matches = eng.match(sig)
     assert matches and matches[0].failure_id == "F-0001"
     assert matches[0].score > 0.98
+
+
+def test_kmeans_gpu_matches_cpu():
+    from kakveda_amd.patterns.kmeans import StreamingKMeans
+
+    g = torch.Generator().manual_seed(21)
+    centers = torch.randn(4, 768, generator=g)
+    centers = centers / centers.norm(dim=-1, keepdim=True)
+    pts = centers.repeat_interleave(64, 0) + 0.05 * torch.randn(256, 768, generator=g)
+    pts = pts / pts.norm(dim=-1, keepdim=True)
+
+    km_cpu = StreamingKMeans(4, 768, device="cpu", seed=5)
+    km_gpu = StreamingKMeans(4, 768, device="cuda", seed=5)
+    a_cpu = km_cpu.fit(pts, iters=10)
+    a_gpu = km_gpu.fit(pts.to("cuda"), iters=10).cpu()
+    torch.cuda.synchronize()
+    # same partition (bf16 rounding can only flip points on blob borders)
+    agree = (a_cpu == a_gpu).float().mean().item()
+    assert agree > 0.98, agree
+    cos = (km_cpu.centroids * km_gpu.centroids.cpu()).sum(-1)
+    assert (cos > 0.99).all(), cos
