@@ -33,6 +33,7 @@ class LocalCluster:
         device: str = "cpu",
         config: Optional[ConfigStore] = None,
         engine: Optional[GfkbEngine] = None,
+        with_dashboard: bool = False,
     ):
         self.tx = Transport()
         cfg = config or ConfigStore()
@@ -70,6 +71,18 @@ class LocalCluster:
             config=cfg,
         )
 
+        self.dashboard = None
+        if with_dashboard:
+            from kakveda_amd.services.dashboard import create_app as dash_create
+
+            self.dashboard = dash_create(
+                data_dir=data_dir,
+                transport=self.tx,
+                urls=u,
+                self_url=u["dashboard"],
+            )
+            self.tx.register_local(u["dashboard"], self.dashboard)
+
         for name, asgi_app in (
             ("event_bus", self.event_bus),
             ("gfkb", self.gfkb),
@@ -86,6 +99,8 @@ class LocalCluster:
         await self.failure_classifier.state.subscribe()
         await self.pattern_detector.state.subscribe()
         await self.health_scoring.state.subscribe()
+        if self.dashboard is not None:
+            await self.dashboard.state.subscribe()
 
     async def warn(self, app_id: str, prompt: str, tools=None, env=None) -> dict:
         resp = await self.tx.post(

@@ -1,0 +1,126 @@
+"""Jinja2 templates for the dashboard HTML pages.
+
+Kept in-package via a DictLoader (parity target: the reference's 24
+template files; here each page is intentionally minimal markup over the
+same data the /api routes expose — the API surface is the contract)."""
+
+from __future__ import annotations
+
+from typing import Any, Dict
+
+from jinja2 import DictLoader, Environment, select_autoescape
+
+_BASE = """<!doctype html><html><head><meta charset="utf-8">
+<title>kakveda-amd</title>
+<style>
+body{font-family:system-ui,sans-serif;margin:2rem;background:#0d1117;color:#e6edf3}
+a{color:#58a6ff} table{border-collapse:collapse;width:100%}
+td,th{border-bottom:1px solid #30363d;padding:.4rem .6rem;text-align:left;font-size:.9rem}
+.nav a{margin-right:1rem} .card{background:#161b22;border:1px solid #30363d;
+border-radius:8px;padding:1rem;margin:1rem 0}
+.bar{height:10px;background:#1f6feb;border-radius:3px}
+.badge{padding:.1rem .5rem;border-radius:1rem;background:#21262d;font-size:.8rem}
+input,textarea,select,button{background:#21262d;color:#e6edf3;border:1px solid #30363d;
+border-radius:6px;padding:.4rem .6rem;margin:.2rem 0}
+button{cursor:pointer;background:#238636}
+</style></head><body>
+<div class="nav"><a href="/">Dashboard</a><a href="/warnings">Warnings</a>
+<a href="/runs">Runs</a><a href="/playground">Playground</a>
+<a href="/agents">Agents</a><a href="/logout">Logout</a></div>
+{% block content %}{% endblock %}
+</body></html>"""
+
+_TEMPLATES = {
+    "base.html": _BASE,
+    "login.html": """{% extends "base.html" %}{% block content %}
+<div class="card"><h2>Sign in</h2>
+<form method="post" action="/login">
+<input name="email" placeholder="email" value="admin@kakveda.local"><br>
+<input name="password" type="password" placeholder="password"><br>
+<button type="submit">Login</button></form>
+<p>Demo users: admin/operator/viewer/demo @kakveda.local</p></div>
+{% endblock %}""",
+    "home.html": """{% extends "base.html" %}{% block content %}
+<h1>Failure Intelligence</h1>
+<p>Signed in as {{ user }} ({{ roles|join(", ") }})</p>
+<div class="card"><h3>Recent failures (GFKB)</h3><table>
+<tr><th>id</th><th>v</th><th>type</th><th>apps</th><th>occurrences</th></tr>
+{% for f in failures %}<tr><td>{{ f.failure_id }}</td><td>{{ f.version }}</td>
+<td>{{ f.failure_type }}</td><td>{{ f.affected_apps|join(", ") }}</td>
+<td>{{ f.occurrences }}</td></tr>{% endfor %}</table></div>
+<div class="card"><h3>Patterns</h3><table>
+<tr><th>id</th><th>name</th><th>apps</th><th>failures</th></tr>
+{% for p in patterns %}<tr><td>{{ p.pattern_id }}</td><td>{{ p.name }}</td>
+<td>{{ p.affected_apps|join(", ") }}</td><td>{{ p.failure_ids|length }}</td></tr>
+{% endfor %}</table></div>
+<div class="card"><h3>Latest warnings</h3><table>
+<tr><th>ts</th><th>app</th><th>action</th><th>confidence</th></tr>
+{% for w in warnings %}<tr><td>{{ w.ts }}</td><td>{{ w.app_id }}</td>
+<td><span class="badge">{{ w.action }}</span></td>
+<td>{{ "%.2f"|format(w.confidence) }}</td></tr>{% endfor %}</table></div>
+<div class="card"><h3>Run a scenario</h3>
+<form method="post" action="/scenarios/run">
+<input name="app_id" value="app-A">
+<input name="prompt" size="60"
+ value="Summarize this and include references even if none are provided.">
+<button type="submit">Run</button></form></div>
+{% endblock %}""",
+    "warnings.html": """{% extends "base.html" %}{% block content %}
+<h1>Warnings</h1>
+<div class="card"><h3>Last {{ analytics.days }} days: {{ analytics.total }} warnings,
+est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
+<table><tr><th>day</th><th>count</th></tr>
+{% for day, n in analytics.daily.items() %}<tr><td>{{ day }}</td><td>{{ n }}</td></tr>{% endfor %}
+</table></div>
+<div class="card"><table>
+<tr><th>ts</th><th>app</th><th>action</th><th>conf</th><th>pattern</th><th>message</th></tr>
+{% for w in warnings %}<tr id="w-{{ w.id }}"><td>{{ w.ts }}</td><td>{{ w.app_id }}</td>
+<td>{{ w.action }}</td><td>{{ "%.2f"|format(w.confidence) }}</td>
+<td>{{ w.pattern_id }}</td><td>{{ w.message[:120] }}</td></tr>{% endfor %}
+</table></div>{% endblock %}""",
+    "runs.html": """{% extends "base.html" %}{% block content %}
+<h1>Runs</h1>
+<form method="get" action="/runs"><input name="q" size="60" value="{{ q }}"
+ placeholder="provider:stub model:llama latency_ms>100 has:error free text">
+<button type="submit">Filter</button></form>
+<div class="card"><table>
+<tr><th>id</th><th>ts</th><th>app</th><th>provider</th><th>model</th>
+<th>latency</th><th>tokens</th><th>cost µUSD</th></tr>
+{% for r in runs %}<tr><td><a href="/runs/{{ r.id }}">{{ r.id }}</a></td>
+<td>{{ r.ts }}</td><td>{{ r.app_id }}</td><td>{{ r.provider }}</td>
+<td>{{ r.model }}</td><td>{{ "%.0f"|format(r.latency_ms) }} ms</td>
+<td>{{ r.tokens_in }}/{{ r.tokens_out }}</td><td>{{ r.cost_usd_micro }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
+    "run_detail.html": """{% extends "base.html" %}{% block content %}
+<h1>Run {{ run.id }} — {{ run.trace_id }}</h1>
+<div class="card"><b>{{ run.provider }}/{{ run.model }}</b> ·
+{{ "%.0f"|format(run.latency_ms) }} ms · {{ run.cost_usd_micro }} µUSD
+<h4>Prompt</h4><pre>{{ prompt }}</pre><h4>Response</h4><pre>{{ response }}</pre></div>
+<div class="card"><h3>Span waterfall</h3>
+{% for sp in spans %}
+<div style="margin-left:{{ sp.depth * 20 }}px">
+<small>{{ sp.name }} — {{ "%.1f"|format(sp.duration_ms) }} ms</small>
+<div class="bar" style="margin-left:{{ sp.pct_left }}%;width:{{ sp.pct_width }}%"></div>
+</div>{% endfor %}</div>{% endblock %}""",
+    "playground.html": """{% extends "base.html" %}{% block content %}
+<h1>Playground</h1>
+<div class="card"><p>POST /api/playground/run with {"prompt", "model",
+"agent_id", "prompt_version_id", "experiment"}.</p>
+<p>Models: {% for m in models %}<span class="badge">{{ m }}</span> {% endfor %}</p>
+</div>{% endblock %}""",
+    "agents.html": """{% extends "base.html" %}{% block content %}
+<h1>Agent registry</h1>
+<div class="card"><table>
+<tr><th>id</th><th>name</th><th>url</th><th>enabled</th><th>capabilities</th>
+<th>heartbeat</th></tr>
+{% for a in agents %}<tr><td>{{ a.id }}</td><td>{{ a.name }}</td>
+<td>{{ a.base_url }}</td><td>{{ a.enabled }}</td>
+<td>{{ a.capabilities|join(", ") }}</td><td>{{ a.last_heartbeat }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
+}
+
+_env = Environment(loader=DictLoader(_TEMPLATES), autoescape=select_autoescape(["html"]))
+
+
+def render(name: str, ctx: Dict[str, Any]) -> str:
+    return _env.get_template(name).render(**ctx)

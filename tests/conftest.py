@@ -1,7 +1,11 @@
 import asyncio
 import inspect
+import os
 
 import pytest
+
+# keep auth tests fast; production default stays 260k iterations
+os.environ.setdefault("KAKVEDA_PBKDF2_ITERS", "1000")
 
 
 def pytest_configure(config):
