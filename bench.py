@@ -59,13 +59,13 @@ def main():
     torch.cuda.set_device(device)
 
     from kakveda_amd import ops
+    from kakveda_amd.parallel.sharded import ShardedStore
 
     if not ops.hip_available():
         raise RuntimeError("HIP extension not built; run __graft_entry__.build() first")
 
     D, B, k = args.dim, args.batch, args.topk
     per_rank = args.entries // world
-    row0 = rank * per_rank  # global id offset of this shard
 
     # ---- build the sharded GFKB (synthetic fingerprints, bf16 in HBM) ----
     gen = torch.Generator(device=device).manual_seed(1000 + rank)
@@ -77,6 +77,8 @@ def main():
             torch.bfloat16
         )
     ops.l2normalize_(corpus)
+    store = ShardedStore(D, device=str(device), capacity=1024)
+    store.load_shard(corpus, per_rank * world)  # zero-copy adopt
 
     # ---- query pool: pre-hashed synthetic signature features --------------
     # (feature hashing is CPU-side and untimed, as in production where the
@@ -95,32 +97,13 @@ def main():
     feat_idx = torch.from_numpy(idx_np).to(device).repeat(reps, 1)[:B].contiguous()
     feat_w = torch.from_numpy(w_np).to(device).repeat(reps, 1)[:B].contiguous()
 
-    if dist:
-        import torch.distributed as td
-
-        gather_buf = [
-            torch.empty(B, 2 * k, dtype=torch.float32, device=device) for _ in range(world)
-        ]
-
     def step() -> int:
-        # 1. encode on GPU
+        # 1. encode on GPU (embedding_bag kernel + projection GEMM + norm)
         q = enc.encode_features(feat_idx, feat_w).to(torch.bfloat16)
-        # 2. fused cosine top-k on this shard
-        scores, lidx = ops.cosine_topk(q, corpus, k)
-        gidx = (lidx + row0).to(torch.float32)
-        if dist:
-            import torch.distributed as td
-
-            # 3. all-gather (score, id) candidates over xGMI
-            packed = torch.cat([scores, gidx], dim=1).contiguous()
-            td.all_gather(gather_buf, packed)
-            all_scores = torch.cat([g[:, :k] for g in gather_buf], dim=1)
-            all_idx = torch.cat([g[:, k:] for g in gather_buf], dim=1)
-            # 4. merge + policy
-            fs, sel = torch.topk(all_scores, k, dim=1)
-            fid = all_idx.gather(1, sel)
-        else:
-            fs, fid = scores, gidx
+        # 2-4. fused cosine top-k per shard, RCCL all-gather over xGMI,
+        #      exact (score, global-id) merge — kakveda_amd.parallel
+        fs, fid = store.search(q, k)
+        # 5. threshold policy decision
         warn = (fs[:, 0] >= args.threshold).sum()
         _ = fid
         return int(warn.item() >= 0)

@@ -65,6 +65,14 @@ class EmbeddingStore:
             fresh[: self.count] = self._data[: self.count]
             self._data = fresh
 
+    def adopt(self, data: torch.Tensor) -> None:
+        """Take ownership of a pre-built [n, D] row tensor (zero copy) —
+        used by benches/restore paths where the rows already live on the
+        device in the right dtype."""
+        assert data.dtype == self.dtype and data.shape[1] == self.dim
+        self._data = data
+        self.count = data.shape[0]
+
     def append(self, rows: torch.Tensor) -> int:
         """Append [n, D] rows; returns the first new row index."""
         n = rows.shape[0]

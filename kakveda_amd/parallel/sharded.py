@@ -51,6 +51,16 @@ class ShardedStore:
         self.total += n
         return first_global
 
+    def load_shard(self, rows: torch.Tensor, total: int) -> None:
+        """Benchmark/restore fast path: install this rank's shard directly.
+        ``rows`` are the rank's round-robin rows in local order (global id
+        of local row l is l*world + rank); ``total`` is the global count."""
+        if rows.dtype == self.local.dtype:
+            self.local.adopt(rows.contiguous())
+        else:
+            self.local.append(rows)
+        self.total = total
+
     def _local_to_global(self, local_idx: torch.Tensor) -> torch.Tensor:
         """Round-robin inverse: local row l on rank r is global l*world + r."""
         out = local_idx * self.world + self.rank
