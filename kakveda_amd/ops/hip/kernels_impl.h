@@ -358,6 +358,281 @@ __global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
 
 inline constexpr auto cosine_topk_partial = cosine_topk_partial_t<0>;
 
+
+// ===========================================================================
+// 256x256-tile counted-pipeline variant (guide T3+T4 adapted).
+//
+// STATUS: EXPERIMENTAL, probe-only (tools/knn_probe.hip modes 4/5); NOT
+// wired into the torch binding. Measured A/B (B=4096 x N=2M x D=768):
+// gemm-only 809 TF vs the 128^2 kernel's 853 TF — this adaptation of the
+// guide's 8-phase template (2-phase windows at BK=32, 3 slots, counted
+// vmcnt(4)) does not reproduce the template's pipelining gains; the
+// finer per-phase ds_read/MFMA/glds interleave appears essential.
+// Kept for the next optimization round.
+//
+// Geometry: block = 512 threads (8 waves as 2 row-halves x 4 col-quads),
+// output tile 256 queries x 256 corpus rows; per-wave output 128x64
+// (acc 8x4 fragments). K advances in 32-deep windows ("K-tiles"), each
+// split into 2 phases (one n-half x all m x 16 MFMA). Staging: 3 LDS
+// slots of (A 16 KiB + B 16 KiB); each wave stages 2 KiB of A and 2 KiB
+// of B per window, issued TWO windows ahead of use, so the per-window
+// `s_waitcnt vmcnt(4)` certifies the incoming K-tile while the next one
+// stays in flight across the RAW barriers (never a vmcnt(0) drain in the
+// main loop). Certification is cross-wave safe because every wave waits
+// vmcnt(4) at the window's first phase and all reads happen after that
+// phase's barrier.
+//
+// A/B images per K-tile: [256 rows][32 k] bf16, 64-byte rows; fragment
+// reads are ds_read_b128 with the 16-B slot swizzled by ((row>>2)&3) on
+// both the glds source and the read (rule 21).
+// ===========================================================================
+
+constexpr int BM2 = 256;
+constexpr int BN2 = 256;
+constexpr int BK2 = 32;
+constexpr int THREADS2 = 512;
+constexpr int IMG_BYTES2 = BM2 * BK2 * 2;       // 16 KiB per operand image
+constexpr int SLOT_BYTES2 = 2 * IMG_BYTES2;     // A+B per K-tile
+constexpr int NSLOT2 = 3;
+
+DEVINL void stage_piece2(const bf16_t* __restrict__ src, int row0, int row_max,
+                         long row_bytes, int ktile_byte, char* img_base,
+                         int piece_off, int lane) {
+  // one 1 KiB piece: wave-uniform LDS base + lane*16; 64-B image rows
+  const int P = piece_off + lane * 16;
+  const int r = P >> 6;
+  const int s_phys = (P >> 4) & 3;
+  const int s_log = s_phys ^ ((r >> 2) & 3);
+  const int gr = min(row0 + r, row_max);
+  const char* gaddr =
+      (const char*)src + (size_t)gr * row_bytes + ktile_byte + s_log * 16;
+  glds16(gaddr, img_base + piece_off);
+}
+
+DEVINL bf16x8 read_frag2(const char* img, int row, int slot) {
+  const int s_phys = slot ^ ((row >> 2) & 3);
+  return *(const bf16x8*)(img + row * 64 + s_phys * 16);
+}
+
+template <int EPI_MODE>  // 0 = full, 1 = GEMM only
+__global__ __launch_bounds__(THREADS2, 2) void cosine_topk_partial256_t(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
+    float* __restrict__ partial_score, int* __restrict__ partial_idx,
+    int B, int N, int D, int chunk_tiles, int nchunks,
+    unsigned* rowthr = nullptr, unsigned long long* stats = nullptr) {
+  __shared__ char smem[NSLOT2 * SLOT_BYTES2 + 2 * BM2 * KMAX * 4];
+  char* const smem0 = smem;
+  auto aimg = [&](int slot) -> char* { return smem0 + slot * SLOT_BYTES2; };
+  auto bimg = [&](int slot) -> char* {
+    return smem0 + slot * SLOT_BYTES2 + IMG_BYTES2;
+  };
+  float* lsc = (float*)(smem + NSLOT2 * SLOT_BYTES2);
+  int* lix = (int*)(smem + NSLOT2 * SLOT_BYTES2 + BM2 * KMAX * 4);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int g = lane >> 4;
+  const int cl = lane & 15;
+  const int wr = wid >> 2;  // row half (0/1): rows wr*128..+128
+  const int wc = wid & 3;   // col quad (0..3): cols wc*64..+64
+
+  // XCD-aware remap (see the 128^2 kernel)
+  const int nrt = gridDim.y;
+  int chunk_id, row_tile;
+  if ((gridDim.x & 7) == 0 && gridDim.x * nrt >= 512) {
+    const int bid = blockIdx.x + gridDim.x * blockIdx.y;
+    const int xcd = bid & 7;
+    const int slot = bid >> 3;
+    const int cpx = gridDim.x >> 3;
+    chunk_id = xcd * cpx + slot / nrt;
+    row_tile = slot % nrt;
+  } else {
+    chunk_id = blockIdx.x;
+    row_tile = blockIdx.y;
+  }
+
+  const int row0 = row_tile * BM2;
+  const long rb = (long)D * 2;
+  const int ntiles_total = (N + BN2 - 1) / BN2;
+  const int tile0 = chunk_id * chunk_tiles;
+  const int tiles_here = min(chunk_tiles, ntiles_total - tile0);
+  const int nkt = D / BK2;  // windows per col tile (24 at D=768)
+
+  for (int i = tid; i < BM2 * KMAX; i += THREADS2) {
+    lsc[i] = NEG_INF;
+    lix[i] = -1;
+  }
+  __syncthreads();
+
+  // stage one window's worth for (global K-tile kt of col tile jt)
+  auto stage_window = [&](int jt, int kt) {
+    if (jt >= tiles_here) return;
+    const int slot = (jt * nkt + kt) % NSLOT2;
+    const int kb = kt * BK2 * 2;
+    const int col0 = (tile0 + jt) * BN2;
+    // A: half(wr) rows wr*128..+128; wave's 2 KiB at + wc*2048
+    const int aoff = wr * 8192 + wc * 2048;
+    stage_piece2(Q, row0, B - 1, rb, kb, aimg(slot), aoff, lane);
+    stage_piece2(Q, row0, B - 1, rb, kb, aimg(slot), aoff + 1024, lane);
+    // B: quad(wc) cols wc*64..+64; wave's 2 KiB at + wr*2048
+    const int boff = wc * 4096 + wr * 2048;
+    stage_piece2(C, col0, N - 1, rb, kb, bimg(slot), boff, lane);
+    stage_piece2(C, col0, N - 1, rb, kb, bimg(slot), boff + 1024, lane);
+  };
+
+  // prologue: first two windows in flight
+  stage_window(0, 0);
+  stage_window(0, min(1, nkt - 1));
+  // NB: nkt >= 2 always (D >= 64)
+
+  for (int j = 0; j < tiles_here; ++j) {
+    const int col0 = (tile0 + j) * BN2;
+
+    float warm = NEG_INF;  // per-lane warm threshold for rows wr*128+{lane, 64+lane}
+    float warm2 = NEG_INF;
+    if (EPI_MODE != 1 && rowthr != nullptr) {
+      const int r1 = row0 + wr * 128 + lane;
+      const int r2 = r1 + 64;
+      if (r1 < B) warm = dec_f32(rowthr[r1]);
+      if (r2 < B) warm2 = dec_f32(rowthr[r2]);
+    }
+
+    f32x4 acc[8][4];
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int kt = 0; kt < nkt; ++kt) {
+      const int slot = (j * nkt + kt) % NSLOT2;
+      const char* Ai = aimg(slot) + wr * 8192;          // wave's A half
+      const char* Bi = bimg(slot) + wc * 4096;          // wave's B quad
+#pragma unroll
+      for (int ph = 0; ph < 2; ++ph) {
+        if (ph == 0) {
+          // certify K-tile kt (its 4 pieces are the oldest outstanding);
+          // K-tile kt+1's 4 glds stay in flight across the raw barrier
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        }
+        {
+          // stage half of the (kt+2) window per phase
+          const int kt2 = kt + 2;
+          const int jt2 = j + kt2 / nkt;
+          const int kk2 = kt2 % nkt;
+          if (ph == 0) {
+            if (jt2 < tiles_here) {
+              const int s2 = (jt2 * nkt + kk2) % NSLOT2;
+              const int kb2 = kk2 * BK2 * 2;
+              const int c2 = (tile0 + jt2) * BN2;
+              const int aoff = wr * 8192 + wc * 2048;
+              stage_piece2(Q, row0, B - 1, rb, kb2, aimg(s2), aoff, lane);
+              stage_piece2(Q, row0, B - 1, rb, kb2, aimg(s2), aoff + 1024, lane);
+              (void)c2;
+            }
+          } else {
+            if (jt2 < tiles_here) {
+              const int s2 = (jt2 * nkt + kk2) % NSLOT2;
+              const int kb2 = kk2 * BK2 * 2;
+              const int c2 = (tile0 + jt2) * BN2;
+              const int boff = wc * 4096 + wr * 2048;
+              stage_piece2(C, c2, N - 1, rb, kb2, bimg(s2), boff, lane);
+              stage_piece2(C, c2, N - 1, rb, kb2, bimg(s2), boff + 1024, lane);
+            }
+          }
+        }
+        __builtin_amdgcn_s_barrier();  // pieces certified for every wave
+        bf16x8 af[8], bf[2];
+#pragma unroll
+        for (int m = 0; m < 8; ++m)
+          af[m] = read_frag2(Ai, m * 16 + cl, g);
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          bf[n] = read_frag2(Bi, (ph * 2 + n) * 16 + cl, g);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int m = 0; m < 8; ++m)
+#pragma unroll
+          for (int n = 0; n < 2; ++n)
+            acc[m][ph * 2 + n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[m], bf[n], acc[m][ph * 2 + n], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_barrier();  // LDS slot reuse guard
+      }
+    }
+
+    // ---- top-k epilogue (same design as the 128^2 kernel) ---------------
+    if constexpr (EPI_MODE == 1) {
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          asm volatile("" ::"v"(acc[m][n]));
+    } else {
+      const int colb = col0 + wc * 64 + cl;
+      // lists are shared across the 4 col-quads of a row: serialise by wc
+#pragma unroll
+      for (int phw = 0; phw < 4; ++phw) {
+        if (wc == phw) {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) {
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+              const int row = wr * 128 + m * 16 + g * 4 + reg;
+              const int rl = m * 16 + g * 4 + reg;  // 0..127 within half
+              const float rwarm =
+                  __shfl(rl < 64 ? warm : warm2, rl & 63, 64);
+              const int lbase = row * KMAX;
+              const float rmin0 = fmaxf(lsc[lbase], rwarm);
+              float w0 = (colb + 0 < N) ? acc[m][0][reg] : NEG_INF;
+              float w1 = (colb + 16 < N) ? acc[m][1][reg] : NEG_INF;
+              float w2 = (colb + 32 < N) ? acc[m][2][reg] : NEG_INF;
+              float w3 = (colb + 48 < N) ? acc[m][3][reg] : NEG_INF;
+              float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+#pragma unroll
+              for (int off = 1; off < 16; off <<= 1)
+                gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
+              if (gmax > rmin0) {
+                topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
+                                         w3, colb, N, lane, g, rowthr,
+                                         (row0 + row < B) ? row0 + row + 1 : 0);
+              }
+            }
+          }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();  // raw: cross-tile glds stay in flight
+      }
+    }
+  }
+
+  // write partials: [B][nchunks][KMAX]
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  if (tid < BM2) {
+    const int grow = row0 + tid;
+    if (grow < B) {
+      const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;
+#pragma unroll
+      for (int q = 0; q < KMAX; ++q) {
+        partial_score[base + q] = lsc[tid * KMAX + q];
+        partial_idx[base + q] = lix[tid * KMAX + q];
+      }
+    }
+  }
+  if (tid >= BM2 && tid < 2 * BM2) {
+    const int grow = row0 + tid;
+    if (grow < B) {
+      const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;
+#pragma unroll
+      for (int q = 0; q < KMAX; ++q) {
+        partial_score[base + q] = lsc[tid * KMAX + q];
+        partial_idx[base + q] = lix[tid * KMAX + q];
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Merge per-chunk partial lists -> final sorted top-k per query row.
 //   grid = B blocks, 256 threads.

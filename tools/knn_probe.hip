@@ -65,6 +65,17 @@ int main(int argc, char** argv) {
   HIP_CHECK(hipMalloc(&rowthr, (size_t)B * 4));
 
   dim3 grid(nchunks, row_tiles);
+  // 256-tile variant geometry
+  const int row_tiles2 = (B + BM2 - 1) / BM2;
+  const int ntiles2 = (int)((N + BN2 - 1) / BN2);
+  long want2 = ((long)ntiles2 * row_tiles2 + 511) / 512;
+  const int chunk_tiles2 = (int)std::max(4L, std::min(want2, 128L));
+  const int nchunks2 = ((ntiles2 + chunk_tiles2 - 1) / chunk_tiles2 + 7) & ~7;
+  float* pscore2;
+  int* pidx2;
+  HIP_CHECK(hipMalloc(&pscore2, (size_t)B * nchunks2 * KMAX * 4));
+  HIP_CHECK(hipMalloc(&pidx2, (size_t)B * nchunks2 * KMAX * 4));
+  dim3 grid2(nchunks2, row_tiles2);
   auto run_mode = [&](int mode) {
     hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0, 0,
                        rowthr, B);
@@ -74,21 +85,33 @@ int main(int argc, char** argv) {
     else if (mode == 1)
       hipLaunchKernelGGL((cosine_topk_partial_t<1>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
-    else
+    else if (mode == 2)
       hipLaunchKernelGGL((cosine_topk_partial_t<2>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 4)
+      hipLaunchKernelGGL((cosine_topk_partial256_t<0>), grid2, dim3(THREADS2), 0, 0,
+                         Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
+    else
+      hipLaunchKernelGGL((cosine_topk_partial256_t<1>), grid2, dim3(THREADS2), 0, 0,
+                         Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
+    hipError_t le = hipGetLastError();
+    if (le != hipSuccess)
+      fprintf(stderr, "launch error (mode %d): %s\n", mode, hipGetErrorString(le));
   };
 
-  for (int m = 0; m < 3; ++m) { run_mode(m); }
+  const int warm_modes[5] = {0, 1, 2, 4, 5};
+  for (int mi = 0; mi < 5; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[3] = {"full", "gemm-only", "precheck+stash"};
-  std::vector<std::vector<float>> ms(3);
+  const char* names[6] = {"full128", "gemm128", "precheck128", "-", "full256", "gemm256"};
+  const int modes[5] = {0, 1, 2, 4, 5};
+  std::vector<std::vector<float>> ms(6);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
   for (int it = 0; it < iters; ++it) {
-    for (int m = 0; m < 3; ++m) {
+    for (int mi = 0; mi < 5; ++mi) {
+      const int m = modes[mi];
       HIP_CHECK(hipEventRecord(t0));
       run_mode(m);
       HIP_CHECK(hipEventRecord(t1));
@@ -114,7 +137,8 @@ int main(int argc, char** argv) {
   printf("stats: stamped=%llu (%.2f/tile) inserts=%llu (%.1f/row/chunk)\n",
          hstats[0], hstats[0] / tiles, hstats[1],
          hstats[1] / ((double)B * nchunks));
-  for (int m = 0; m < 3; ++m) {
+  for (int mi = 0; mi < 5; ++mi) {
+    const int m = modes[mi];
     std::sort(ms[m].begin(), ms[m].end());
     const float med = ms[m][ms[m].size() / 2];
     const float mn = ms[m][0];
