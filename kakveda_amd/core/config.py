@@ -47,7 +47,8 @@ class ConfigStore:
     def _load(self, force: bool = False) -> None:
         now = time.monotonic()
         hot = self._data.get("hot_reload", {}) or {}
-        poll = float(hot.get("poll_seconds", 2) or 2)
+        raw_poll = hot.get("poll_seconds", 2)
+        poll = 2.0 if raw_poll is None else float(raw_poll)  # 0 = poll always
         if not force:
             if not bool(hot.get("enabled", True)):
                 return
