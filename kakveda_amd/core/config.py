@@ -8,6 +8,7 @@ governed by ``hot_reload.poll_seconds`` in the file itself.
 
 from __future__ import annotations
 
+import copy
 import os
 import time
 from pathlib import Path
@@ -39,7 +40,7 @@ class ConfigStore:
 
     def __init__(self, path: Optional[str] = None):
         self.path = Path(path or os.environ.get("CONFIG_PATH", "/app/config/config.yaml"))
-        self._data: Dict[str, Any] = dict(DEFAULT_CONFIG)
+        self._data: Dict[str, Any] = copy.deepcopy(DEFAULT_CONFIG)
         self._mtime: float = -1.0
         self._last_check: float = 0.0
         self._load(force=True)
@@ -66,7 +67,8 @@ class ConfigStore:
         except Exception:
             return
         if isinstance(loaded, dict):
-            merged = dict(DEFAULT_CONFIG)
+            merged = copy.deepcopy(DEFAULT_CONFIG)  # deep: _deep_update must
+            # never mutate the shared defaults
             _deep_update(merged, loaded)
             self._data = merged
             self._mtime = mtime
