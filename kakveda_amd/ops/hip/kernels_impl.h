@@ -96,6 +96,56 @@ DEVINL bf16x8 read_frag(const char* lds_tile, int row, int slot) {
 // so the 16 unrolled epilogue bodies don't get their candidate values
 // hoisted live simultaneously (measured: inlining costs 100+ VGPRs and
 // ~850 B/lane of scratch spill).
+// Ballot-leader variant: no per-insert wave argmax reduce — each loop
+// iteration picks the lowest qualifying lane, which inserts its own best
+// candidate. Fewer cross-lane ops per insert; insertion order is
+// arbitrary (list semantics identical).
+template <bool PUBLISH>
+__device__ __noinline__ void topk_extract_group_bl(
+    volatile float* vsc, volatile int* vix, int lbase, float rwarm,
+    float w0, float w1, float w2, float w3, int colb, int N_unused, int lane,
+    int g, unsigned* rowthr, int growp1) {
+  float rmin = fmaxf(vsc[lbase], rwarm);
+  while (true) {
+    float b = w0;
+    int bn = 0;
+    if (w1 > b) { b = w1; bn = 1; }
+    if (w2 > b) { b = w2; bn = 2; }
+    if (w3 > b) { b = w3; bn = 3; }
+    const unsigned long long ball =
+        __ballot(b > rmin) & (0xFFFFull << (g * 16));
+    if (!ball) break;
+    const int leader = __ffsll(ball) - 1;
+    if (lane == leader) {
+      float nmn = b;
+      int nmp = 0;
+#pragma unroll
+      for (int q = 1; q < KMAX; ++q) {
+        const float s = vsc[lbase + q];
+        if (s < nmn) { nmn = s; nmp = q; }
+      }
+      const int gcol = colb + bn * 16;
+      if (nmp == 0) {
+        vsc[lbase] = b;
+        vix[lbase] = gcol;
+      } else {
+        const int mi2 = vix[lbase + nmp];
+        vsc[lbase] = nmn;
+        vix[lbase] = mi2;
+        vsc[lbase + nmp] = b;
+        vix[lbase + nmp] = gcol;
+      }
+      if (bn == 0) w0 = NEG_INF;
+      else if (bn == 1) w1 = NEG_INF;
+      else if (bn == 2) w2 = NEG_INF;
+      else w3 = NEG_INF;
+      if (PUBLISH && rowthr != nullptr && nmn > NEG_INF && growp1 > 0)
+        atomicMax(&rowthr[growp1 - 1], enc_f32(nmn));
+    }
+    rmin = fmaxf(vsc[lbase], rwarm);  // same-wave LDS program order
+  }
+}
+
 template <bool PUBLISH>
 __device__ __noinline__ void topk_extract_group(
     volatile float* vsc, volatile int* vix, int lbase, float rwarm,
@@ -307,7 +357,11 @@ __global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
             if constexpr (EPI_MODE == 3) {
               if (cl == 0 && stats) atomicAdd(&stats[0], 1ull);
             }
-            if constexpr (EPI_MODE != 2) {
+            if constexpr (EPI_MODE == 6) {
+              topk_extract_group_bl<true>(lsc, lix, lbase, rwarm, w0, w1,
+                                          w2, w3, colb, N, lane, g, rowthr,
+                                          (row0 + row < B) ? row0 + row + 1 : 0);
+            } else if constexpr (EPI_MODE != 2) {
               topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
                                        w3, colb, N, lane, g, rowthr,
                                        (row0 + row < B) ? row0 + row + 1 : 0);

@@ -66,8 +66,13 @@ def main() -> None:
 
     import uvicorn
 
+    from kakveda_amd.core.otel import instrument_fastapi, setup_otel
+
+    app = build_app(args.service)
+    setup_otel(f"kakveda-{args.service.replace('_', '-')}")
+    instrument_fastapi(app)
     uvicorn.run(
-        build_app(args.service),
+        app,
         host=args.host,
         port=args.port or DEFAULT_PORTS[args.service],
         log_level="info",
