@@ -91,6 +91,33 @@ DEVINL bf16x8 read_frag(const char* lds_tile, int row, int slot) {
   return *(const bf16x8*)(lds_tile + row * 128 + s_phys * 16);
 }
 
+// Generalised variants: tiles of [128 rows][NKK*32 k] (row = NKK*64 bytes).
+template <int NKK>
+DEVINL void stage_tile_n(const bf16_t* __restrict__ src, int row0, int row_max,
+                         long row_bytes, int ktile_byte, char* lds_tile,
+                         int wid, int lane) {
+  constexpr int ROWB = NKK * 64;     // bytes per tile row
+  constexpr int SLOTS = NKK * 4;     // 16-B slots per row
+#pragma unroll
+  for (int i = 0; i < 2 * NKK; ++i) {
+    const int lds_off = wid * (2 * NKK) * 1024 + i * 1024;
+    const int P = lds_off + lane * 16;
+    const int r = P / ROWB;
+    const int s_phys = (P >> 4) % SLOTS;
+    const int s_log = NKK == 2 ? (s_phys ^ (r & 7)) : (s_phys ^ ((r >> 2) & 3));
+    const int gr = min(row0 + r, row_max);
+    const char* gaddr =
+        (const char*)src + (size_t)gr * row_bytes + ktile_byte + s_log * 16;
+    glds16(gaddr, lds_tile + lds_off);
+  }
+}
+
+template <int NKK>
+DEVINL bf16x8 read_frag_n(const char* lds_tile, int row, int slot) {
+  const int s_phys = NKK == 2 ? (slot ^ (row & 7)) : (slot ^ ((row >> 2) & 3));
+  return *(const bf16x8*)(lds_tile + row * (NKK * 64) + s_phys * 16);
+}
+
 
 // Extraction for one qualifying (row, col-half): deliberately __noinline__
 // so the 16 unrolled epilogue bodies don't get their candidate values
@@ -201,8 +228,8 @@ __device__ __noinline__ void topk_extract_group(
 //   grid.x = nchunks, grid.y = ceil(B/128), block = 256 threads.
 //   partial_score/partial_idx: [B][nchunks][KMAX]
 // ---------------------------------------------------------------------------
-template <int EPI_MODE>  // 0 = full, 1 = GEMM only, 2 = pre-check only, 3 = full + stats
-__global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
+template <int EPI_MODE, int NKK = 2>  // NKK: 32-deep K steps per LDS stage (2 -> BK=64)
+__global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
     float* __restrict__ partial_score, int* __restrict__ partial_idx,
     int B, int N, int D, int chunk_tiles, int nchunks,
@@ -211,14 +238,15 @@ __global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
   // private top-k lists. 80 KiB total -> 2 blocks/CU. The lists are
   // private to the one wave that computes that (row-half, col-half), so
   // the epilogue needs NO barriers and overlaps the next tile's staging.
-  __shared__ char smem[2 * TILE_BYTES * 2 + 2 * BM * KMAX * 8];
+  constexpr int TB = BM * NKK * 32 * 2;  // one staged tile
+  __shared__ char smem[2 * TB * 2 + 2 * BM * KMAX * 8];
   char* const smem0 = smem;  // avoid static-init addrspacecast of arrays
-  auto abuf = [&](int i) -> char* { return smem0 + i * TILE_BYTES; };
-  auto bbuf = [&](int i) -> char* { return smem0 + (2 + i) * TILE_BYTES; };
+  auto abuf = [&](int i) -> char* { return smem0 + i * TB; };
+  auto bbuf = [&](int i) -> char* { return smem0 + (2 + i) * TB; };
   // list layout: scores[wc][row][KMAX] then idx[wc][row][KMAX];
   // invariant: slot 0 of each list holds that list's MINIMUM.
-  float* lsc = (float*)(smem + 4 * TILE_BYTES);
-  int* lix = (int*)(smem + 4 * TILE_BYTES + 2 * BM * KMAX * 4);
+  float* lsc = (float*)(smem + 4 * TB);
+  int* lix = (int*)(smem + 4 * TB + 2 * BM * KMAX * 4);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -254,7 +282,7 @@ __global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
   const int ntiles_total = (N + BN - 1) / BN;
   const int tile0 = chunk_id * chunk_tiles;
   const int tiles_here = min(chunk_tiles, ntiles_total - tile0);
-  const int nkt = D / BK;
+  const int nkt = D / (NKK * 32);
 
   // init lists (-inf scores; slot 0 is the min by construction)
   for (int i = tid; i < 2 * BM * KMAX; i += THREADS) {
@@ -292,23 +320,23 @@ __global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
       // tile j+1's first K-tile, so the epilogue below runs while that
       // staging is in flight.
       if (kt + 1 < nkt) {
-        const int kb = (kt + 1) * BK * 2;
-        stage_tile(Q, row0, B - 1, qrow_bytes, kb, abuf(cur ^ 1), wid, lane);
-        stage_tile(C, col0, N - 1, qrow_bytes, kb, bbuf(cur ^ 1), wid, lane);
+        const int kb = (kt + 1) * NKK * 64;
+        stage_tile_n<NKK>(Q, row0, B - 1, qrow_bytes, kb, abuf(cur ^ 1), wid, lane);
+        stage_tile_n<NKK>(C, col0, N - 1, qrow_bytes, kb, bbuf(cur ^ 1), wid, lane);
       } else if (j + 1 < tiles_here) {
-        stage_tile(Q, row0, B - 1, qrow_bytes, 0, abuf(cur ^ 1), wid, lane);
-        stage_tile(C, col0 + BN, N - 1, qrow_bytes, 0, bbuf(cur ^ 1), wid, lane);
+        stage_tile_n<NKK>(Q, row0, B - 1, qrow_bytes, 0, abuf(cur ^ 1), wid, lane);
+        stage_tile_n<NKK>(C, col0 + BN, N - 1, qrow_bytes, 0, bbuf(cur ^ 1), wid, lane);
       }
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
+      for (int kk = 0; kk < NKK; ++kk) {
         bf16x8 afrag[4], bfrag[4];
         const int slot = kk * 4 + g;
 #pragma unroll
         for (int m = 0; m < 4; ++m)
-          afrag[m] = read_frag(abuf(cur), wr * 64 + m * 16 + cl, slot);
+          afrag[m] = read_frag_n<NKK>(abuf(cur), wr * 64 + m * 16 + cl, slot);
 #pragma unroll
         for (int n = 0; n < 4; ++n)
-          bfrag[n] = read_frag(bbuf(cur), wc * 64 + n * 16 + cl, slot);
+          bfrag[n] = read_frag_n<NKK>(bbuf(cur), wc * 64 + n * 16 + cl, slot);
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -410,7 +438,7 @@ __global__ __launch_bounds__(THREADS, 2) void cosine_topk_partial_t(
   }
 }
 
-inline constexpr auto cosine_topk_partial = cosine_topk_partial_t<0>;
+inline constexpr auto cosine_topk_partial = cosine_topk_partial_t<0, 2>;
 
 
 // ===========================================================================

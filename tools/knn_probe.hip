@@ -88,6 +88,12 @@ int main(int argc, char** argv) {
     else if (mode == 2)
       hipLaunchKernelGGL((cosine_topk_partial_t<2>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 7)
+      hipLaunchKernelGGL((cosine_topk_partial_t<0, 1>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 8)
+      hipLaunchKernelGGL((cosine_topk_partial_t<1, 1>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
@@ -102,13 +108,13 @@ int main(int argc, char** argv) {
       fprintf(stderr, "launch error (mode %d): %s\n", mode, hipGetErrorString(le));
   };
 
-  const int warm_modes[5] = {0, 1, 2, 6, 5};
+  const int warm_modes[5] = {0, 1, 7, 8, 5};
   for (int mi = 0; mi < 5; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[7] = {"full128", "gemm128", "precheck128", "-", "full256", "gemm256", "full128-bl"};
-  const int modes[5] = {0, 1, 2, 6, 5};
-  std::vector<std::vector<float>> ms(7);
+  const char* names[9] = {"full128", "gemm128", "precheck128", "-", "full256", "gemm256", "full128-bl", "full128-bk32", "gemm128-bk32"};
+  const int modes[5] = {0, 1, 7, 8, 5};
+  std::vector<std::vector<float>> ms(9);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
