@@ -38,7 +38,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--entries", type=int, default=10_000_000)
-    p.add_argument("--batch", type=int, default=4096)
+    p.add_argument("--batch", type=int, default=2048)
     p.add_argument("--topk", type=int, default=5)
     p.add_argument("--dim", type=int, default=768)
     p.add_argument("--threshold", type=float, default=0.8)
