@@ -584,6 +584,23 @@ def create_app(
             },
         )
 
+    @app.get("/health", response_class=HTMLResponse)
+    async def health_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        apps: Dict[str, List[Dict]] = {}
+        with ctx.Session() as s:
+            app_ids = [r[0] for r in s.query(dbm.TraceRun.app_id).distinct().limit(20)]
+        for app_id in app_ids:
+            try:
+                resp = await ctx.tx.get(f"{u['health_scoring']}/health/{app_id}", params={"limit": 5})
+                pts = resp.json().get("points", [])
+                if pts:
+                    apps[app_id] = pts
+            except Exception:
+                continue
+        return render("health.html", {"apps": apps})
+
     @app.get("/warnings", response_class=HTMLResponse)
     async def warnings_page(request: Request):
         if not ctx.current_user(request):

@@ -462,6 +462,14 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
                 ]
             }
 
+    @app.get("/datasets", response_class=HTMLResponse)
+    async def datasets_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        ds = await list_datasets()
+        ev = await list_evals()
+        return render("datasets.html", {"datasets": ds["datasets"], "evals": ev["evals"]})
+
     # ======================================================================
     # prompts
     # ======================================================================
@@ -536,6 +544,13 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
                 ]
             }
 
+    @app.get("/prompts", response_class=HTMLResponse)
+    async def prompts_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        data = await list_prompts()
+        return render("prompts.html", {"prompts": data["prompts"]})
+
     # ======================================================================
     # experiments
     # ======================================================================
@@ -585,6 +600,13 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
             s.add(e)
             s.commit()
             return {"ok": True, "id": e.id}
+
+    @app.get("/experiments", response_class=HTMLResponse)
+    async def experiments_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        data = await list_experiments()
+        return render("experiments.html", {"experiments": data["experiments"]})
 
     # ======================================================================
     # agent registry

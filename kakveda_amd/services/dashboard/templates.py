@@ -26,7 +26,9 @@ button{cursor:pointer;background:#238636}
 </style></head><body>
 <div class="nav"><a href="/">Dashboard</a><a href="/warnings">Warnings</a>
 <a href="/runs">Runs</a><a href="/playground">Playground</a>
-<a href="/agents">Agents</a><a href="/logout">Logout</a></div>
+<a href="/agents">Agents</a><a href="/datasets">Datasets</a>
+<a href="/prompts">Prompts</a><a href="/experiments">Experiments</a>
+<a href="/health">Health</a><a href="/logout">Logout</a></div>
 {% block content %}{% endblock %}
 </body></html>"""
 
@@ -117,6 +119,42 @@ est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
 <td>{{ a.base_url }}</td><td>{{ a.enabled }}</td>
 <td>{{ a.capabilities|join(", ") }}</td><td>{{ a.last_heartbeat }}</td></tr>
 {% endfor %}</table></div>{% endblock %}""",
+    "datasets.html": """{% extends "base.html" %}{% block content %}
+<h1>Datasets & evals</h1>
+<div class="card"><table><tr><th>id</th><th>name</th><th>examples</th></tr>
+{% for d in datasets %}<tr><td>{{ d.id }}</td><td>{{ d.name }}</td>
+<td>{{ d.examples }}</td></tr>{% endfor %}</table></div>
+<div class="card"><h3>Evaluation runs</h3>
+<table><tr><th>id</th><th>ts</th><th>name</th><th>pass rate</th><th>p50/p95 ms</th></tr>
+{% for e in evals %}<tr><td>{{ e.id }}</td><td>{{ e.ts }}</td><td>{{ e.name }}</td>
+<td>{{ "%.0f%%"|format(100 * e.summary.get("pass_rate", 0)) }}</td>
+<td>{{ "%.0f"|format(e.summary.get("p50_ms", 0)) }}/{{ "%.0f"|format(e.summary.get("p95_ms", 0)) }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
+    "prompts.html": """{% extends "base.html" %}{% block content %}
+<h1>Prompt library</h1>
+<div class="card"><table>
+<tr><th>id</th><th>name</th><th>latest v</th><th>default model</th><th>tags</th></tr>
+{% for p in prompts %}<tr><td>{{ p.id }}</td><td>{{ p.name }}</td>
+<td>v{{ p.latest_version }}</td><td>{{ p.default_model }}</td><td>{{ p.tags }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
+    "experiments.html": """{% extends "base.html" %}{% block content %}
+<h1>Experiments</h1>
+<div class="card"><table>
+<tr><th>id</th><th>name</th><th>runs</th><th>p50/p95 ms</th><th>providers</th><th>cost µUSD</th></tr>
+{% for e in experiments %}<tr><td>{{ e.id }}</td><td>{{ e.name }}</td><td>{{ e.runs }}</td>
+<td>{{ "%.0f"|format(e.p50_ms) }}/{{ "%.0f"|format(e.p95_ms) }}</td>
+<td>{% for k, v in e.providers.items() %}{{ k }}:{{ v }} {% endfor %}</td>
+<td>{{ e.cost_usd_micro }}</td></tr>{% endfor %}</table></div>{% endblock %}""",
+    "health.html": """{% extends "base.html" %}{% block content %}
+<h1>App health</h1>
+{% for app_id, points in apps.items() %}
+<div class="card"><h3>{{ app_id }}</h3><table>
+<tr><th>ts</th><th>score</th><th>failure rate</th><th>penalty</th></tr>
+{% for p in points %}<tr><td>{{ p.ts }}</td><td>{{ "%.1f"|format(p.score) }}</td>
+<td>{{ "%.2f"|format(p.failure_rate) }}</td><td>{{ p.recurrent_penalty }}</td></tr>
+{% endfor %}</table></div>
+{% else %}<div class="card">No health points yet — ingest failures or use
+POST /health/test.</div>{% endfor %}{% endblock %}""",
 }
 
 _env = Environment(loader=DictLoader(_TEMPLATES), autoescape=select_autoescape(["html"]))

@@ -281,7 +281,8 @@ async def test_html_pages(tmp_path):
         await _login(client)
         home = await client.get("/")
         assert "Failure Intelligence" in home.text
-        for page in ("/warnings", "/runs", "/playground", "/agents"):
+        for page in ("/warnings", "/runs", "/playground", "/agents",
+                     "/datasets", "/prompts", "/experiments", "/health"):
             resp = await client.get(page)
             assert resp.status_code == 200, page
     await cluster.aclose()
