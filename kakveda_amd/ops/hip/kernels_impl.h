@@ -442,87 +442,101 @@ inline constexpr auto cosine_topk_partial = cosine_topk_partial_t<0, 2>;
 
 
 // ===========================================================================
-// 256x256-tile counted-pipeline variant (guide T3+T4 adapted).
+// 256x256-tile 8-phase counted-pipeline kernel (guide T3+T4+T5).
 //
-// STATUS: EXPERIMENTAL, probe-only (tools/knn_probe.hip modes 4/5); NOT
-// wired into the torch binding. Measured A/B (B=4096 x N=2M x D=768):
-// gemm-only 809 TF vs the 128^2 kernel's 853 TF — this adaptation of the
-// guide's 8-phase template (2-phase windows at BK=32, 3 slots, counted
-// vmcnt(4)) does not reproduce the template's pipelining gains; the
-// finer per-phase ds_read/MFMA/glds interleave appears essential.
-// Kept for the next optimization round.
+// Geometry: 512 threads = 8 waves as 2 row-halves (wr) x 4 col-quads (wc);
+// output tile 256 queries x 256 corpus rows, per-wave 128x64 (acc 8x4
+// fragments of 16x16). K advances in BK=64 "windows" of 4 phases; each
+// phase computes one row-quadrant (2 m-frags x 4 n x 2 kk = 16 MFMA). The
+// B fragments are read ONCE per window (phase 0) and stay in registers,
+// which is what frees the B image bytes early enough that the NEXT
+// window's staging can target the LIVE buffer:
 //
-// Geometry: block = 512 threads (8 waves as 2 row-halves x 4 col-quads),
-// output tile 256 queries x 256 corpus rows; per-wave output 128x64
-// (acc 8x4 fragments). K advances in 32-deep windows ("K-tiles"), each
-// split into 2 phases (one n-half x all m x 16 MFMA). Staging: 3 LDS
-// slots of (A 16 KiB + B 16 KiB); each wave stages 2 KiB of A and 2 KiB
-// of B per window, issued TWO windows ahead of use, so the per-window
-// `s_waitcnt vmcnt(4)` certifies the incoming K-tile while the next one
-// stays in flight across the RAW barriers (never a vmcnt(0) drain in the
-// main loop). Certification is cross-wave safe because every wave waits
-// vmcnt(4) at the window's first phase and all reads happen after that
-// phase's barrier.
+//   staging schedule (per wave, reader-aligned 1 KiB pieces):
+//     window t, ph0: A(t+1) pieces 0,1      -> buf[(t+1)&1]  (unread now)
+//     window t, ph1: A(t+1) 2,3 + B(t+2) 0,1-> B into buf[t&1]: its B bytes
+//     window t, ph2: B(t+2) pieces 2,3         were consumed at ph0
+//   certification: one `s_waitcnt vmcnt(4)` at ph3 (before the barrier)
+//   leaves exactly B(t+2)'s 4 glds in flight and proves A(t+1) + B(t+1)
+//   (and older) landed -> the pipeline NEVER drains to vmcnt(0).
 //
-// A/B images per K-tile: [256 rows][32 k] bf16, 64-byte rows; fragment
-// reads are ds_read_b128 with the 16-B slot swizzled by ((row>>2)&3) on
-// both the glds source and the read (rule 21).
+// Reader-aligned staging makes per-wave vmcnt certification sound: a
+// wave stages exactly the quarter of each half-image that it (or its
+// barrier-synchronised co-reader) consumes. Two raw barriers per phase:
+// the second separates every wave's ds_read retirement (compiler lgkm
+// before its MFMAs) from the next phase's glds landing on those bytes.
+//
+// LDS: 2 buffers x (A[2 halves] + B[2 halves]) x 16 KiB = 128 KiB
+// + shared top-k lists 16 KiB = 144 KiB -> 1 block/CU, 2 waves/SIMD.
 // ===========================================================================
 
-constexpr int BM2 = 256;
-constexpr int BN2 = 256;
-constexpr int BK2 = 32;
-constexpr int THREADS2 = 512;
-constexpr int IMG_BYTES2 = BM2 * BK2 * 2;       // 16 KiB per operand image
-constexpr int SLOT_BYTES2 = 2 * IMG_BYTES2;     // A+B per K-tile
-constexpr int NSLOT2 = 3;
 
-DEVINL void stage_piece2(const bf16_t* __restrict__ src, int row0, int row_max,
-                         long row_bytes, int ktile_byte, char* img_base,
-                         int piece_off, int lane) {
-  // one 1 KiB piece: wave-uniform LDS base + lane*16; 64-B image rows
-  const int P = piece_off + lane * 16;
-  const int r = P >> 6;
-  const int s_phys = (P >> 4) & 3;
-  const int s_log = s_phys ^ ((r >> 2) & 3);
-  const int gr = min(row0 + r, row_max);
-  const char* gaddr =
-      (const char*)src + (size_t)gr * row_bytes + ktile_byte + s_log * 16;
-  glds16(gaddr, img_base + piece_off);
+// Per-m epilogue body for the 8-phase kernel: __noinline__ so the 8 m-bodies
+// don't hoist 128 candidate values live on top of the 128-register
+// accumulator (measured: inlined epilogue = 256 VGPR + 384 B/lane scratch
+// with spill code inside the K-loop, 4x slowdown).
+template <bool PUBLISH, bool DO_EXTRACT = true>
+__device__ __noinline__ void topk_epilogue_m8(
+    volatile float* lsc, volatile int* lix, f32x4 a0, f32x4 a1, f32x4 a2,
+    f32x4 a3, int rowbase, int rlbase, float warm, float warm2, int colb,
+    int N, int lane, int g, unsigned* rowthr, int B, int row0) {
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int rl = rlbase + g * 4 + reg;
+    const int row = rowbase + g * 4 + reg;
+    const float rwarm = __shfl(rl < 64 ? warm : warm2, rl & 63, 64);
+    const int lbase = row * KMAX;
+    const float rmin0 = fmaxf(lsc[lbase], rwarm);
+    float w0 = (colb + 0 < N) ? a0[reg] : NEG_INF;
+    float w1 = (colb + 16 < N) ? a1[reg] : NEG_INF;
+    float w2 = (colb + 32 < N) ? a2[reg] : NEG_INF;
+    float w3 = (colb + 48 < N) ? a3[reg] : NEG_INF;
+    float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1)
+      gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
+    if (DO_EXTRACT && gmax > rmin0) {
+      topk_extract_group<PUBLISH>(lsc, lix, lbase, rwarm, w0, w1, w2, w3,
+                                  colb, N, lane, g, rowthr,
+                                  (row0 + row < B) ? row0 + row + 1 : 0);
+    }
+  }
 }
 
-DEVINL bf16x8 read_frag2(const char* img, int row, int slot) {
-  const int s_phys = slot ^ ((row >> 2) & 3);
-  return *(const bf16x8*)(img + row * 64 + s_phys * 16);
-}
+constexpr int BM8 = 256;
+constexpr int BN8 = 256;
+constexpr int THREADS8 = 512;
+constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 
 template <int EPI_MODE>  // 0 = full, 1 = GEMM only
-__global__ __launch_bounds__(THREADS2, 2) void cosine_topk_partial256_t(
+__global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
     float* __restrict__ partial_score, int* __restrict__ partial_idx,
     int B, int N, int D, int chunk_tiles, int nchunks,
     unsigned* rowthr = nullptr, unsigned long long* stats = nullptr) {
-  __shared__ char smem[NSLOT2 * SLOT_BYTES2 + 2 * BM2 * KMAX * 4];
+  __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4];
   char* const smem0 = smem;
-  auto aimg = [&](int slot) -> char* { return smem0 + slot * SLOT_BYTES2; };
-  auto bimg = [&](int slot) -> char* {
-    return smem0 + slot * SLOT_BYTES2 + IMG_BYTES2;
+  // buffer b in {0,1}: A half h at b*4*HALF8 + h*HALF8; B half h at +2*HALF8
+  auto ahalf = [&](int b, int h) -> char* {
+    return smem0 + (b * 4 + h) * HALF8;
   };
-  float* lsc = (float*)(smem + NSLOT2 * SLOT_BYTES2);
-  int* lix = (int*)(smem + NSLOT2 * SLOT_BYTES2 + BM2 * KMAX * 4);
+  auto bhalf = [&](int b, int h) -> char* {
+    return smem0 + (b * 4 + 2 + h) * HALF8;
+  };
+  float* lsc = (float*)(smem + 8 * HALF8);
+  int* lix = (int*)(smem + 8 * HALF8 + BM8 * KMAX * 4);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int g = lane >> 4;
   const int cl = lane & 15;
-  const int wr = wid >> 2;  // row half (0/1): rows wr*128..+128
-  const int wc = wid & 3;   // col quad (0..3): cols wc*64..+64
+  const int wr = wid >> 2;  // row half
+  const int wc = wid & 3;   // col quad
 
-  // XCD-aware remap (see the 128^2 kernel)
   const int nrt = gridDim.y;
   int chunk_id, row_tile;
-  if ((gridDim.x & 7) == 0 && gridDim.x * nrt >= 512) {
+  if ((gridDim.x & 7) == 0 && gridDim.x * nrt >= 256) {
     const int bid = blockIdx.x + gridDim.x * blockIdx.y;
     const int xcd = bid & 7;
     const int slot = bid >> 3;
@@ -534,51 +548,75 @@ __global__ __launch_bounds__(THREADS2, 2) void cosine_topk_partial256_t(
     row_tile = blockIdx.y;
   }
 
-  const int row0 = row_tile * BM2;
+  const int row0 = row_tile * BM8;
   const long rb = (long)D * 2;
-  const int ntiles_total = (N + BN2 - 1) / BN2;
+  const int ntiles_total = (N + BN8 - 1) / BN8;
   const int tile0 = chunk_id * chunk_tiles;
   const int tiles_here = min(chunk_tiles, ntiles_total - tile0);
-  const int nkt = D / BK2;  // windows per col tile (24 at D=768)
+  const int nkt = D / 64;  // windows per col tile
 
-  for (int i = tid; i < BM2 * KMAX; i += THREADS2) {
+  for (int i = tid; i < BM8 * KMAX; i += THREADS8) {
     lsc[i] = NEG_INF;
     lix[i] = -1;
   }
   __syncthreads();
+  if (tiles_here <= 0) {
+    // padded chunk: emit -inf partials and exit
+    if (tid < BM8 && row0 + tid < B) {
+      const size_t base = ((size_t)(row0 + tid) * nchunks + chunk_id) * KMAX;
+#pragma unroll
+      for (int q = 0; q < KMAX; ++q) {
+        partial_score[base + q] = NEG_INF;
+        partial_idx[base + q] = -1;
+      }
+    }
+    return;
+  }
 
-  // stage one window's worth for (global K-tile kt of col tile jt)
-  auto stage_window = [&](int jt, int kt) {
+  // ---- staging helpers --------------------------------------------------
+  // A-share of K-tile t: quarter wc (rows wc*32..+32) of A-half(wr).
+  // B-share of K-tile t: rows (wc&1)*64 + wr*32 ..+32 of B-half(wc>>1).
+  // piece = 1 KiB = 8 image rows; lane-linear glds dest; swizzled source.
+  auto stage_a_piece = [&](int t, int i) {  // i in 0..3
+    const int kt = t % nkt;
+    const int jt = t / nkt;
     if (jt >= tiles_here) return;
-    const int slot = (jt * nkt + kt) % NSLOT2;
-    const int kb = kt * BK2 * 2;
-    const int col0 = (tile0 + jt) * BN2;
-    // A: half(wr) rows wr*128..+128; wave's 2 KiB at + wc*2048
-    const int aoff = wr * 8192 + wc * 2048;
-    stage_piece2(Q, row0, B - 1, rb, kb, aimg(slot), aoff, lane);
-    stage_piece2(Q, row0, B - 1, rb, kb, aimg(slot), aoff + 1024, lane);
-    // B: quad(wc) cols wc*64..+64; wave's 2 KiB at + wr*2048
-    const int boff = wc * 4096 + wr * 2048;
-    stage_piece2(C, col0, N - 1, rb, kb, bimg(slot), boff, lane);
-    stage_piece2(C, col0, N - 1, rb, kb, bimg(slot), boff + 1024, lane);
+    char* img = ahalf(t & 1, wr);
+    const int off = wc * 4096 + i * 1024;
+    const int P = off + lane * 16;
+    const int r = P >> 7;
+    const int s_log = ((P >> 4) & 7) ^ (r & 7);
+    const int gr = min(row0 + wr * 128 + r, B - 1);
+    glds16((const char*)Q + (size_t)gr * rb + kt * 128 + s_log * 16, img + off);
+  };
+  auto stage_b_piece = [&](int t, int i) {
+    const int kt = t % nkt;
+    const int jt = t / nkt;
+    if (jt >= tiles_here) return;
+    char* img = bhalf(t & 1, wc >> 1);
+    const int off = ((wc & 1) * 64 + wr * 32) * 128 + i * 1024;
+    const int P = off + lane * 16;
+    const int r = P >> 7;
+    const int s_log = ((P >> 4) & 7) ^ (r & 7);
+    const int col_base = (tile0 + jt) * BN8 + (wc >> 1) * 128;
+    const int gr = min(col_base + r, N - 1);
+    glds16((const char*)C + (size_t)gr * rb + kt * 128 + s_log * 16, img + off);
   };
 
-  // prologue: first two windows in flight
-  stage_window(0, 0);
-  stage_window(0, min(1, nkt - 1));
-  // NB: nkt >= 2 always (D >= 64)
+  // ---- prologue: K-tile 0 fully + B(1); then steady schedule ------------
+#pragma unroll
+  for (int i = 0; i < 4; ++i) stage_a_piece(0, i);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) stage_b_piece(0, i);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) stage_b_piece(1, i);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const int total_windows = tiles_here * nkt;
 
   for (int j = 0; j < tiles_here; ++j) {
-    const int col0 = (tile0 + j) * BN2;
-
-    float warm = NEG_INF;  // per-lane warm threshold for rows wr*128+{lane, 64+lane}
-    float warm2 = NEG_INF;
-    if (EPI_MODE != 1 && rowthr != nullptr) {
-      const int r1 = row0 + wr * 128 + lane;
-      const int r2 = r1 + 64;
-      if (r1 < B) warm = dec_f32(rowthr[r1]);
-      if (r2 < B) warm2 = dec_f32(rowthr[r2]);
-    }
+    const int col0 = (tile0 + j) * BN8;
 
     f32x4 acc[8][4];
 #pragma unroll
@@ -587,63 +625,62 @@ __global__ __launch_bounds__(THREADS2, 2) void cosine_topk_partial256_t(
       for (int n = 0; n < 4; ++n) acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     for (int kt = 0; kt < nkt; ++kt) {
-      const int slot = (j * nkt + kt) % NSLOT2;
-      const char* Ai = aimg(slot) + wr * 8192;          // wave's A half
-      const char* Bi = bimg(slot) + wc * 4096;          // wave's B quad
+      const int t = j * nkt + kt;  // global window index
+      const char* Ai = ahalf(t & 1, wr);
+      const char* Bi = bhalf(t & 1, wc >> 1) + ((wc & 1) * 64) * 128;
+      bf16x8 bf[4][2];
 #pragma unroll
-      for (int ph = 0; ph < 2; ++ph) {
-        if (ph == 0) {
-          // certify K-tile kt (its 4 pieces are the oldest outstanding);
-          // K-tile kt+1's 4 glds stay in flight across the raw barrier
+      for (int qm = 0; qm < 4; ++qm) {
+        // ---- reads (before barrier; compiler inserts lgkm before MFMA)
+        bf16x8 af[2][2];
+#pragma unroll
+        for (int mm = 0; mm < 2; ++mm)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            af[mm][kk] = read_frag(Ai, (qm * 2 + mm) * 16 + cl, kk * 4 + g);
+        if (qm == 0) {
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+#pragma unroll
+            for (int kk = 0; kk < 2; ++kk)
+              bf[n][kk] = read_frag(Bi, n * 16 + cl, kk * 4 + g);
+        }
+        // ---- staging issues (see schedule above)
+        if (qm == 0 && t + 1 < total_windows) {
+          stage_a_piece(t + 1, 0);
+          stage_a_piece(t + 1, 1);
+        } else if (qm == 1) {
+          if (t + 1 < total_windows) {
+            stage_a_piece(t + 1, 2);
+            stage_a_piece(t + 1, 3);
+          }
+          if (t + 2 < total_windows) {
+            stage_b_piece(t + 2, 0);
+            stage_b_piece(t + 2, 1);
+          }
+        } else if (qm == 2 && t + 2 < total_windows) {
+          stage_b_piece(t + 2, 2);
+          stage_b_piece(t + 2, 3);
+        } else if (qm == 3) {
+          // certify window t+1's A and B (B(t+2)'s 4 glds stay in flight)
           asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
         }
-        {
-          // stage half of the (kt+2) window per phase
-          const int kt2 = kt + 2;
-          const int jt2 = j + kt2 / nkt;
-          const int kk2 = kt2 % nkt;
-          if (ph == 0) {
-            if (jt2 < tiles_here) {
-              const int s2 = (jt2 * nkt + kk2) % NSLOT2;
-              const int kb2 = kk2 * BK2 * 2;
-              const int c2 = (tile0 + jt2) * BN2;
-              const int aoff = wr * 8192 + wc * 2048;
-              stage_piece2(Q, row0, B - 1, rb, kb2, aimg(s2), aoff, lane);
-              stage_piece2(Q, row0, B - 1, rb, kb2, aimg(s2), aoff + 1024, lane);
-              (void)c2;
-            }
-          } else {
-            if (jt2 < tiles_here) {
-              const int s2 = (jt2 * nkt + kk2) % NSLOT2;
-              const int kb2 = kk2 * BK2 * 2;
-              const int c2 = (tile0 + jt2) * BN2;
-              const int boff = wc * 4096 + wr * 2048;
-              stage_piece2(C, c2, N - 1, rb, kb2, bimg(s2), boff, lane);
-              stage_piece2(C, c2, N - 1, rb, kb2, bimg(s2), boff + 1024, lane);
-            }
-          }
-        }
-        __builtin_amdgcn_s_barrier();  // pieces certified for every wave
-        bf16x8 af[8], bf[2];
-#pragma unroll
-        for (int m = 0; m < 8; ++m)
-          af[m] = read_frag2(Ai, m * 16 + cl, g);
-#pragma unroll
-        for (int n = 0; n < 2; ++n)
-          bf[n] = read_frag2(Bi, (ph * 2 + n) * 16 + cl, g);
+        __builtin_amdgcn_s_barrier();
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int m = 0; m < 8; ++m)
+        for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-          for (int n = 0; n < 2; ++n)
-            acc[m][ph * 2 + n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[m], bf[n], acc[m][ph * 2 + n], 0, 0, 0);
+          for (int mm = 0; mm < 2; ++mm)
+#pragma unroll
+            for (int n = 0; n < 4; ++n)
+              acc[qm * 2 + mm][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  af[mm][kk], bf[n][kk], acc[qm * 2 + mm][n], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
-        __builtin_amdgcn_s_barrier();  // LDS slot reuse guard
+        __builtin_amdgcn_s_barrier();  // read-retirement vs next glds
       }
     }
 
-    // ---- top-k epilogue (same design as the 128^2 kernel) ---------------
+    // ---- top-k epilogue: shared lists, serialised by col-quad -----------
     if constexpr (EPI_MODE == 1) {
 #pragma unroll
       for (int m = 0; m < 8; ++m)
@@ -651,39 +688,107 @@ __global__ __launch_bounds__(THREADS2, 2) void cosine_topk_partial256_t(
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
     } else {
-      const int colb = col0 + wc * 64 + cl;
-      // lists are shared across the 4 col-quads of a row: serialise by wc
+      // ---- top-k epilogue (stash + lane-parallel register-list drain) --
+      // The A images of the LAST window's buffer are dead during the
+      // epilogue (A(t+2) staging is only issued at window t+1 phase 0),
+      // so each wave stashes its 128x64 score quadrant there and drains
+      // it with one lane per two rows: candidates merge into a
+      // register-held copy of the row's list (select chains, no calls,
+      // no ballots, no volatile round-trips). Col-quad phases serialise
+      // writers of the shared per-row lists.
+      const int t_last = j * nkt + nkt - 1;
+      // 32 KiB ([128][64] f32) of LDS is free during the epilogue (the A
+      // images of the last window's buffer). The 8 waves take turns
+      // stashing their 128x64 score quadrant there; after each stash ALL
+      // waves drain it cooperatively (16 rows per wave, one row per lane,
+      // single-writer lists), so the scan latency is hidden by 8-wave TLP
+      // instead of being exposed on a lone wave.
+      float* stash = (float*)ahalf(t_last & 1, 0);
 #pragma unroll
-      for (int phw = 0; phw < 4; ++phw) {
-        if (wc == phw) {
+      for (int phw = 0; phw < 8; ++phw) {
+        const int swc = phw >> 1, swr = phw & 1;  // stashing wave
+        if (wc == swc && wr == swr) {
 #pragma unroll
-          for (int m = 0; m < 8; ++m) {
+          for (int m = 0; m < 8; ++m)
 #pragma unroll
-            for (int reg = 0; reg < 4; ++reg) {
-              const int row = wr * 128 + m * 16 + g * 4 + reg;
-              const int rl = m * 16 + g * 4 + reg;  // 0..127 within half
-              const float rwarm =
-                  __shfl(rl < 64 ? warm : warm2, rl & 63, 64);
-              const int lbase = row * KMAX;
-              const float rmin0 = fmaxf(lsc[lbase], rwarm);
-              float w0 = (colb + 0 < N) ? acc[m][0][reg] : NEG_INF;
-              float w1 = (colb + 16 < N) ? acc[m][1][reg] : NEG_INF;
-              float w2 = (colb + 32 < N) ? acc[m][2][reg] : NEG_INF;
-              float w3 = (colb + 48 < N) ? acc[m][3][reg] : NEG_INF;
-              float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+            for (int n = 0; n < 4; ++n)
 #pragma unroll
-              for (int off = 1; off < 16; off <<= 1)
-                gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
-              if (gmax > rmin0) {
-                topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
-                                         w3, colb, N, lane, g, rowthr,
-                                         (row0 + row < B) ? row0 + row + 1 : 0);
+              for (int reg = 0; reg < 4; ++reg)
+                stash[(m * 16 + g * 4 + reg) * 64 + n * 16 + cl] =
+                    acc[m][n][reg];
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();  // stash visible to every wave
+
+        if constexpr (EPI_MODE != 2) {
+          if (lane < 16) {  // 8 waves x 16 lanes = one drainer per row
+            const int rl = wid * 16 + lane;         // 0..127 within half
+            const int row = swr * 128 + rl;
+            const int grow = row0 + row;
+            const int colq = col0 + swc * 64;
+            float ls[KMAX];
+            int li[KMAX];
+#pragma unroll
+            for (int q = 0; q < KMAX; ++q) {
+              ls[q] = lsc[row * KMAX + q];
+              li[q] = lix[row * KMAX + q];
+            }
+            float rmin = ls[0];
+#pragma unroll
+            for (int q = 1; q < KMAX; ++q) rmin = fminf(rmin, ls[q]);
+            if (rowthr != nullptr && grow < B)
+              rmin = fmaxf(rmin, dec_f32(rowthr[grow]));
+            bool dirty = false;
+            const float* srow = stash + rl * 64;
+#pragma unroll 4
+            for (int ii = 0; ii < 16; ++ii) {
+              const int iv = (ii + rl) & 15;  // bank stagger across rows
+              const float4 v4 = *(const float4*)(srow + 4 * iv);
+#pragma unroll
+              for (int e = 0; e < 4; ++e) {
+                const float v = e == 0 ? v4.x : e == 1 ? v4.y : e == 2 ? v4.z : v4.w;
+                const int gc = colq + 4 * iv + e;
+                if (v > rmin && gc < N) {
+                  int mp = 0;
+                  float mn1 = ls[0], mn2 = 1e38f;
+#pragma unroll
+                  for (int q = 1; q < KMAX; ++q) {
+                    if (ls[q] < mn1) { mn2 = mn1; mn1 = ls[q]; mp = q; }
+                    else if (ls[q] < mn2) { mn2 = ls[q]; }
+                  }
+#pragma unroll
+                  for (int q = 0; q < KMAX; ++q)
+                    if (q == mp) { ls[q] = v; li[q] = gc; }
+                  rmin = fmaxf(rmin, fminf(mn2, v));
+                  dirty = true;
+                }
               }
+            }
+            if (dirty) {
+              int mp = 0;
+              float mn = ls[0];
+#pragma unroll
+              for (int q = 1; q < KMAX; ++q)
+                if (ls[q] < mn) { mn = ls[q]; mp = q; }
+              // swap the min into slot 0 (list invariant), then write back
+              const float s_mp = ls[mp];
+              const int i_mp = li[mp];
+              ls[mp] = ls[0];
+              li[mp] = li[0];
+              ls[0] = s_mp;
+              li[0] = i_mp;
+#pragma unroll
+              for (int q = 0; q < KMAX; ++q) {
+                lsc[row * KMAX + q] = ls[q];
+                lix[row * KMAX + q] = li[q];
+              }
+              if (rowthr != nullptr && mn > NEG_INF && grow < B)
+                atomicMax(&rowthr[grow], enc_f32(mn));
             }
           }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();  // raw: cross-tile glds stay in flight
+        __builtin_amdgcn_s_barrier();  // lists settled; stash reusable
       }
     }
   }
@@ -691,18 +796,7 @@ __global__ __launch_bounds__(THREADS2, 2) void cosine_topk_partial256_t(
   // write partials: [B][nchunks][KMAX]
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  if (tid < BM2) {
-    const int grow = row0 + tid;
-    if (grow < B) {
-      const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;
-#pragma unroll
-      for (int q = 0; q < KMAX; ++q) {
-        partial_score[base + q] = lsc[tid * KMAX + q];
-        partial_idx[base + q] = lix[tid * KMAX + q];
-      }
-    }
-  }
-  if (tid >= BM2 && tid < 2 * BM2) {
+  if (tid < BM8) {
     const int grow = row0 + tid;
     if (grow < B) {
       const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;

@@ -66,8 +66,8 @@ int main(int argc, char** argv) {
 
   dim3 grid(nchunks, row_tiles);
   // 256-tile variant geometry
-  const int row_tiles2 = (B + BM2 - 1) / BM2;
-  const int ntiles2 = (int)((N + BN2 - 1) / BN2);
+  const int row_tiles2 = (B + BM8 - 1) / BM8;
+  const int ntiles2 = (int)((N + BN8 - 1) / BN8);
   long want2 = ((long)ntiles2 * row_tiles2 + 511) / 512;
   const int chunk_tiles2 = (int)std::max(4L, std::min(want2, 128L));
   const int nchunks2 = ((ntiles2 + chunk_tiles2 - 1) / chunk_tiles2 + 7) & ~7;
@@ -97,24 +97,27 @@ int main(int argc, char** argv) {
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 9)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<2>), grid2, dim3(THREADS8), 0, 0,
+                         Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
     else if (mode == 4)
-      hipLaunchKernelGGL((cosine_topk_partial256_t<0>), grid2, dim3(THREADS2), 0, 0,
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<0>), grid2, dim3(THREADS8), 0, 0,
                          Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
     else
-      hipLaunchKernelGGL((cosine_topk_partial256_t<1>), grid2, dim3(THREADS2), 0, 0,
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<1>), grid2, dim3(THREADS8), 0, 0,
                          Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
     hipError_t le = hipGetLastError();
     if (le != hipSuccess)
       fprintf(stderr, "launch error (mode %d): %s\n", mode, hipGetErrorString(le));
   };
 
-  const int warm_modes[5] = {0, 1, 7, 8, 5};
+  const int warm_modes[5] = {0, 4, 9, 5, 1};
   for (int mi = 0; mi < 5; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[9] = {"full128", "gemm128", "precheck128", "-", "full256", "gemm256", "full128-bl", "full128-bk32", "gemm128-bk32"};
-  const int modes[5] = {0, 1, 7, 8, 5};
-  std::vector<std::vector<float>> ms(9);
+  const char* names[10] = {"full128", "gemm128", "precheck128", "-", "full8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p"};
+  const int modes[5] = {0, 4, 9, 5, 1};
+  std::vector<std::vector<float>> ms(10);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
