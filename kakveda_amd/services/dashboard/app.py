@@ -487,6 +487,8 @@ def create_app(
             s.commit()
             warning_id = we.id
         audit(ctx, (payload or {}).get("sub", "anonymous"), "scenario.run", app_id)
+        if "text/html" in request.headers.get("accept", ""):
+            return RedirectResponse(f"/warnings#w-{warning_id}", status_code=303)
         return {
             "ok": True,
             "trace_id": trace_id,
@@ -600,6 +602,34 @@ def create_app(
             except Exception:
                 continue
         return render("health.html", {"apps": apps})
+
+    @app.get("/failure/{fid}", response_class=HTMLResponse)
+    async def failure_detail(request: Request, fid: str):
+        """Failure detail with version addressing: /failure/F-0001 shows the
+        latest version, /failure/F-0001v3 a specific one (reference
+        dashboard/app.py:1822-1909)."""
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        base_id, _, ver = fid.partition("v")
+        try:
+            resp = await ctx.tx.get(f"{u['gfkb']}/failures")
+            records = [
+                r for r in resp.json().get("failures", [])
+                if r.get("failure_id") == base_id
+            ]
+        except Exception:
+            records = []
+        if not records:
+            return HTMLResponse("failure not found", status_code=404)
+        if ver:
+            chosen = [r for r in records if str(r.get("version")) == ver]
+            record = chosen[0] if chosen else records[-1]
+        else:
+            record = records[-1]
+        return render(
+            "failure_detail.html",
+            {"record": record, "versions": records, "fid": base_id},
+        )
 
     @app.get("/warnings", response_class=HTMLResponse)
     async def warnings_page(request: Request):

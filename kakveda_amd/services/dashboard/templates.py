@@ -119,6 +119,23 @@ est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
 <td>{{ a.base_url }}</td><td>{{ a.enabled }}</td>
 <td>{{ a.capabilities|join(", ") }}</td><td>{{ a.last_heartbeat }}</td></tr>
 {% endfor %}</table></div>{% endblock %}""",
+    "failure_detail.html": """{% extends "base.html" %}{% block content %}
+<h1>{{ record.failure_id }} v{{ record.version }} — {{ record.failure_type }}</h1>
+<div class="card">
+<p><b>Severity:</b> {{ record.impact_severity }} ·
+<b>Occurrences:</b> {{ record.occurrences }} ·
+<b>Apps:</b> {{ record.affected_apps|join(", ") }}</p>
+<p><b>Root cause:</b> {{ record.root_cause }}</p>
+<p><b>Resolution:</b> {{ record.resolution }}</p>
+<p><b>Signature:</b> <code>{{ record.signature_text }}</code></p>
+</div>
+<div class="card"><h3>Version history</h3><table>
+<tr><th>version</th><th>updated</th><th>occurrences</th><th>apps</th></tr>
+{% for v in versions %}<tr>
+<td><a href="/failure/{{ fid }}v{{ v.version }}">v{{ v.version }}</a></td>
+<td>{{ v.updated_at }}</td><td>{{ v.occurrences }}</td>
+<td>{{ v.affected_apps|join(", ") }}</td></tr>{% endfor %}
+</table></div>{% endblock %}""",
     "datasets.html": """{% extends "base.html" %}{% block content %}
 <h1>Datasets & evals</h1>
 <div class="card"><table><tr><th>id</th><th>name</th><th>examples</th></tr>

@@ -286,3 +286,20 @@ async def test_html_pages(tmp_path):
             resp = await client.get(page)
             assert resp.status_code == 200, page
     await cluster.aclose()
+
+
+async def test_failure_detail_version_addressing(tmp_path):
+    cluster = await _cluster(tmp_path)
+    async with _client(cluster) as client:
+        await _login(client)
+        # create two versions of the same failure identity
+        await client.post("/scenarios/run", json={"app_id": "app-A", "prompt": PROMPT})
+        await client.post("/scenarios/run", json={"app_id": "app-B", "prompt": PROMPT})
+        latest = await client.get("/failure/F-0001")
+        assert latest.status_code == 200
+        assert "v2" in latest.text and "HALLUCINATION_CITATION" in latest.text
+        v1 = await client.get("/failure/F-0001v1")
+        assert "v1" in v1.text
+        missing = await client.get("/failure/F-9999")
+        assert missing.status_code == 404
+    await cluster.aclose()
