@@ -662,8 +662,14 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
           stage_b_piece(t + 2, 2);
           stage_b_piece(t + 2, 3);
         } else if (qm == 3) {
-          // certify window t+1's A and B (B(t+2)'s 4 glds stay in flight)
-          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+          // certify window t+1's A and B. Steady state: the newest 4
+          // outstanding glds are B(t+2)'s, so vmcnt(4) proves A(t+1) and
+          // B(t+1) landed. At the chunk tail no B(t+2) was issued, so the
+          // newest 4 would be A(t+1) itself -> drain fully there (rare).
+          if (t + 2 < total_windows)
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+          else
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_s_barrier();
         __builtin_amdgcn_s_setprio(1);
