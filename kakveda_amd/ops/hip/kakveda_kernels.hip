@@ -111,9 +111,38 @@ torch::Tensor embedding_bag(torch::Tensor table, torch::Tensor idx, torch::Tenso
   return out;
 }
 
+std::tuple<torch::Tensor, torch::Tensor> kmeans_update(
+    torch::Tensor points, torch::Tensor assign, int64_t n_clusters) {
+  check_bf16_2d(points, "points");
+  TORCH_CHECK(assign.is_cuda() && assign.scalar_type() == torch::kInt32 &&
+                  assign.is_contiguous(),
+              "assign must be contiguous i32 on GPU");
+  const int N = points.size(0);
+  const int D = points.size(1);
+  const int C = (int)n_clusters;
+  TORCH_CHECK(D % 64 == 0, "D must be a multiple of 64");
+  TORCH_CHECK(C >= 1 && C <= 256, "n_clusters must be in [1,256]");
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(points.device());
+  auto sums = torch::zeros({C, D}, opts);
+  auto counts = torch::zeros({C}, opts);
+  // chunk so the grid oversubscribes the CUs
+  const int dim_tiles = D / 64;
+  int chunks = std::max(1, 2048 / dim_tiles);
+  const int ppc = (N + chunks - 1) / chunks;
+  chunks = (N + ppc - 1) / ppc;
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(kmeans_update_kernel, dim3(dim_tiles, chunks), dim3(256),
+                     C * 64 * 4, stream.stream(),
+                     (const bf16_t*)points.data_ptr(), assign.data_ptr<int>(),
+                     sums.data_ptr<float>(), counts.data_ptr<float>(), N, D, C,
+                     ppc);
+  return {sums, counts};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cosine_topk", &cosine_topk, "fused cosine top-k (MFMA + LDS top-k)");
   m.def("l2normalize_", &l2normalize_, "in-place row L2 normalisation");
   m.def("embedding_bag", &embedding_bag, "weighted embedding bag");
+  m.def("kmeans_update", &kmeans_update, "segmented centroid sum + counts");
   m.attr("KMAX") = KMAX;
 }

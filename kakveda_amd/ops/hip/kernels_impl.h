@@ -816,6 +816,52 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 }
 
 // ---------------------------------------------------------------------------
+// k-means centroid update: segmented reduction of bf16 points into
+// per-cluster fp32 sums + counts. Grid: (dim-tiles, point-chunks); each
+// block accumulates a [C][64]-dim LDS partial over its point chunk, then
+// adds it to the global sums once (C <= 256; LDS = C*64*4 <= 64 KiB).
+// Points are read exactly once per dim-tile (coalesced 128-B row slices);
+// bandwidth-bound by design, replacing torch index_add_ (which was ~30x
+// over the I/O floor in benchmarks/kmeans_bench.py).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void kmeans_update_kernel(
+    const bf16_t* __restrict__ points, const int* __restrict__ assign,
+    float* __restrict__ sums, float* __restrict__ counts, int N, int D,
+    int C, int points_per_chunk) {
+  extern __shared__ float part[];  // [C][64]
+  const int tid = threadIdx.x;
+  const int d0 = blockIdx.x * 64;
+  const int p0 = blockIdx.y * points_per_chunk;
+  const int pend = min(p0 + points_per_chunk, N);
+
+  for (int i = tid; i < C * 64; i += 256) part[i] = 0.f;
+  __syncthreads();
+
+  // 4 points in flight per pass: thread (pl, d) = (tid>>6, tid&63)
+  const int pl = tid >> 6;
+  const int d = tid & 63;
+  for (int p = p0 + pl; p < pend; p += 4) {
+    const int a = assign[p];
+    const float v = (float)points[(size_t)p * D + d0 + d];
+    if (a >= 0 && a < C) atomicAdd(&part[a * 64 + d], v);
+  }
+  __syncthreads();
+
+  for (int i = tid; i < C * 64; i += 256) {
+    const float v = part[i];
+    if (v != 0.f)
+      atomicAdd(&sums[(size_t)(i / 64) * D + d0 + i % 64], v);
+  }
+  // counts: once (first dim-tile only), one pass over the chunk
+  if (blockIdx.x == 0) {
+    for (int p = p0 + tid; p < pend; p += 256) {
+      const int a = assign[p];
+      if (a >= 0 && a < C) atomicAdd(&counts[a], 1.0f);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Merge per-chunk partial lists -> final sorted top-k per query row.
 //   grid = B blocks, 256 threads.
 // ---------------------------------------------------------------------------

@@ -182,3 +182,25 @@ def test_kmeans_gpu_matches_cpu():
     assert agree > 0.98, agree
     cos = (km_cpu.centroids * km_gpu.centroids.cpu()).sum(-1)
     assert (cos > 0.99).all(), cos
+
+
+def test_kmeans_update_kernel_vs_torch():
+    from kakveda_amd import ops
+
+    g = torch.Generator().manual_seed(31)
+    N, D, C = 20000, 768, 64
+    pts = torch.randn(N, D, generator=g).to(_dev(), torch.bfloat16)
+    assign = torch.randint(0, C, (N,), generator=g, dtype=torch.int32).to(_dev())
+
+    sums, counts = ops.kmeans_update(pts, assign, C)
+    torch.cuda.synchronize()
+
+    ref_s = torch.zeros(C, D, device=_dev())
+    ref_s.index_add_(0, assign.long(), pts.float())
+    ref_c = torch.zeros(C, device=_dev())
+    ref_c.index_add_(0, assign.long(), torch.ones(N, device=_dev()))
+
+    assert torch.equal(counts, ref_c)
+    assert torch.allclose(sums, ref_s, atol=0.5, rtol=1e-2), (
+        (sums - ref_s).abs().max().item()
+    )
