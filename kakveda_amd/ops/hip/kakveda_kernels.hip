@@ -59,7 +59,14 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   dim3 grid(nchunks, row_tiles);
   hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0,
                      stream.stream(), (unsigned*)rowthr.data_ptr<int>(), B);
-  if (use8p) {
+  if (k == 1 && !use8p) {
+    // assignment fast path: per-row argmax epilogue (no lists/extraction)
+    hipLaunchKernelGGL((cosine_topk_partial_t<4>), grid, dim3(THREADS), 0, stream.stream(),
+                       (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       B, N, D, chunk_tiles, nchunks,
+                       (unsigned*)nullptr, (unsigned long long*)nullptr);
+  } else if (use8p) {
     hipLaunchKernelGGL((cosine_topk_partial8p_t<0>), grid, dim3(THREADS8), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
@@ -72,10 +79,19 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
   }
-  hipLaunchKernelGGL(topk_merge, dim3(B), dim3(THREADS), 0, stream.stream(),
-                     pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                     out_score.data_ptr<float>(), (long*)out_idx.data_ptr<int64_t>(),
-                     nchunks, (int)k);
+  if (nchunks * KMAX <= 1024) {
+    // few candidates per row: one thread per row beats the block-per-row
+    // merge (whose thread-0 serial scan dominates at large B)
+    hipLaunchKernelGGL(topk_merge_small, dim3((B + 255) / 256), dim3(256), 0,
+                       stream.stream(), pscore.data_ptr<float>(),
+                       pidx.data_ptr<int>(), out_score.data_ptr<float>(),
+                       (long*)out_idx.data_ptr<int64_t>(), B, nchunks, (int)k);
+  } else {
+    hipLaunchKernelGGL(topk_merge, dim3(B), dim3(THREADS), 0, stream.stream(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       out_score.data_ptr<float>(), (long*)out_idx.data_ptr<int64_t>(),
+                       nchunks, (int)k);
+  }
   return {out_score, out_idx};
 }
 
@@ -132,7 +148,7 @@ std::tuple<torch::Tensor, torch::Tensor> kmeans_update(
   chunks = (N + ppc - 1) / ppc;
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(kmeans_update_kernel, dim3(dim_tiles, chunks), dim3(256),
-                     C * 64 * 4, stream.stream(),
+                     C * 65 * 4, stream.stream(),
                      (const bf16_t*)points.data_ptr(), assign.data_ptr<int>(),
                      sums.data_ptr<float>(), counts.data_ptr<float>(), N, D, C,
                      ppc);

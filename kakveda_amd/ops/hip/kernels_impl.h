@@ -228,6 +228,8 @@ __device__ __noinline__ void topk_extract_group(
 //   grid.x = nchunks, grid.y = ceil(B/128), block = 256 threads.
 //   partial_score/partial_idx: [B][nchunks][KMAX]
 // ---------------------------------------------------------------------------
+// EPI_MODE: 0 full top-k, 1 GEMM-only ablation, 2 pre-check only,
+// 3 full+stats, 4 argmax (k=1 fast path: per-row max, no lists/extraction)
 template <int EPI_MODE, int NKK = 2>  // NKK: 32-deep K steps per LDS stage (2 -> BK=64)
 __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
@@ -385,7 +387,26 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
             if constexpr (EPI_MODE == 3) {
               if (cl == 0 && stats) atomicAdd(&stats[0], 1ull);
             }
-            if constexpr (EPI_MODE == 6) {
+            if constexpr (EPI_MODE == 4) {
+              // k=1: group argmax, leader max-merges into list slot 0
+              float b = w0;
+              int bn = 0;
+              if (w1 > b) { b = w1; bn = 1; }
+              if (w2 > b) { b = w2; bn = 2; }
+              if (w3 > b) { b = w3; bn = 3; }
+              float mv = b;
+              int mlane = lane;
+#pragma unroll
+              for (int off = 1; off < 16; off <<= 1) {
+                const float ov = __shfl_xor(mv, off, 64);
+                const int ol = __shfl_xor(mlane, off, 64);
+                if (ov > mv || (ov == mv && ol < mlane)) { mv = ov; mlane = ol; }
+              }
+              if (lane == mlane && mv > lsc[lbase]) {
+                lsc[lbase] = mv;
+                lix[lbase] = colb + bn * 16;
+              }
+            } else if constexpr (EPI_MODE == 6) {
               topk_extract_group_bl<true>(lsc, lix, lbase, rwarm, w0, w1,
                                           w2, w3, colb, N, lane, g, rowthr,
                                           (row0 + row < B) ? row0 + row + 1 : 0);
@@ -828,13 +849,14 @@ __global__ __launch_bounds__(256) void kmeans_update_kernel(
     const bf16_t* __restrict__ points, const int* __restrict__ assign,
     float* __restrict__ sums, float* __restrict__ counts, int N, int D,
     int C, int points_per_chunk) {
-  extern __shared__ float part[];  // [C][64]
+  extern __shared__ float part[];  // [C][64] sums + [C] counts
   const int tid = threadIdx.x;
   const int d0 = blockIdx.x * 64;
   const int p0 = blockIdx.y * points_per_chunk;
   const int pend = min(p0 + points_per_chunk, N);
 
-  for (int i = tid; i < C * 64; i += 256) part[i] = 0.f;
+  float* cpart = part + C * 64;
+  for (int i = tid; i < C * 65; i += 256) part[i] = 0.f;
   __syncthreads();
 
   // 4 points in flight per pass: thread (pl, d) = (tid>>6, tid&63)
@@ -852,12 +874,16 @@ __global__ __launch_bounds__(256) void kmeans_update_kernel(
     if (v != 0.f)
       atomicAdd(&sums[(size_t)(i / 64) * D + d0 + i % 64], v);
   }
-  // counts: once (first dim-tile only), one pass over the chunk
+  // counts: LDS partial first (the naive global-atomic version costs ~1M
+  // contended RMWs per batch), then C adds per chunk
   if (blockIdx.x == 0) {
     for (int p = p0 + tid; p < pend; p += 256) {
       const int a = assign[p];
-      if (a >= 0 && a < C) atomicAdd(&counts[a], 1.0f);
+      if (a >= 0 && a < C) atomicAdd(&cpart[a], 1.0f);
     }
+    __syncthreads();
+    for (int i = tid; i < C; i += 256)
+      if (cpart[i] != 0.f) atomicAdd(&counts[i], cpart[i]);
   }
 }
 
@@ -939,6 +965,56 @@ __global__ __launch_bounds__(THREADS) void topk_merge(
       out_score[(size_t)row * k + q] = fs[q] <= NEG_INF ? -INFINITY : fs[q];
       out_idx[(size_t)row * k + q] = fi[q];
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Small-candidate merge: one THREAD per query row (the block-per-row merge
+// above serialises a 2048-entry scan on thread 0, which dominates when B
+// is large and nchunks small — e.g. k-means assignment at B=1M, C=64).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void topk_merge_small(
+    const float* __restrict__ partial_score, const int* __restrict__ partial_idx,
+    float* __restrict__ out_score, long* __restrict__ out_idx, int B,
+    int nchunks, int k) {
+  const int row = blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= B) return;
+  const int total = nchunks * KMAX;
+  const size_t base = (size_t)row * total;
+  float ls[KMAX];
+  int li[KMAX];
+#pragma unroll
+  for (int q = 0; q < KMAX; ++q) { ls[q] = NEG_INF; li[q] = -1; }
+  float lmin = NEG_INF;
+  for (int e = 0; e < total; ++e) {
+    const float s = partial_score[base + e];
+    if (s > lmin) {
+      int mp = 0;
+      float mn1 = ls[0], mn2 = 1e38f;
+#pragma unroll
+      for (int q = 1; q < KMAX; ++q) {
+        if (ls[q] < mn1) { mn2 = mn1; mn1 = ls[q]; mp = q; }
+        else if (ls[q] < mn2) { mn2 = ls[q]; }
+      }
+#pragma unroll
+      for (int q = 0; q < KMAX; ++q)
+        if (q == mp) { ls[q] = s; li[q] = partial_idx[base + e]; }
+      lmin = fminf(mn2, s);
+    }
+  }
+  // sort descending (insertion sort over KMAX)
+#pragma unroll
+  for (int a = 1; a < KMAX; ++a) {
+    const float s = ls[a];
+    const int ix = li[a];
+    int b = a - 1;
+    for (; b >= 0 && ls[b] < s; --b) { ls[b + 1] = ls[b]; li[b + 1] = li[b]; }
+    ls[b + 1] = s;
+    li[b + 1] = ix;
+  }
+  for (int q = 0; q < k; ++q) {
+    out_score[(size_t)row * k + q] = ls[q] <= NEG_INF ? -INFINITY : ls[q];
+    out_idx[(size_t)row * k + q] = li[q];
   }
 }
 

@@ -55,9 +55,21 @@ class StreamingKMeans:
 
         Distributed: every rank passes its own points; centroid partials
         are all-reduced so all ranks hold identical centroids after.
+        The quality metric reuses the assignment's top-1 scores (free)
+        instead of re-gathering centroids per point.
         """
         pts = points.to(self.device)
-        assign = self.assign(pts)
+        if self.device.type == "cuda":
+            scores, idx = ops.cosine_topk(
+                pts.to(torch.bfloat16), self._centroids_matcher(), 1, self.k
+            )
+            assign = idx[:, 0]
+            mean_cos = scores[:, 0].mean()
+        else:
+            sims = pts.float() @ self.centroids.t()
+            top = sims.max(dim=1)
+            assign = top.indices
+            mean_cos = top.values.mean()
         sums, counts = ops.kmeans_update(
             pts.float() if self.device.type == "cpu" else pts.to(torch.bfloat16),
             assign,
@@ -78,8 +90,7 @@ class StreamingKMeans:
         self.centroids = new / norm
         self._counts = self.decay * self._counts + counts
 
-        sims = (pts.float() * self.centroids[assign].float()).sum(dim=-1)
-        return assign, float(sims.mean().item())
+        return assign, float(mean_cos.item())
 
     def fit(self, points: torch.Tensor, iters: int = 10) -> torch.Tensor:
         """Full-batch Lloyd iterations (decay ignored: hard reassignment)."""
