@@ -59,6 +59,11 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   dim3 grid(nchunks, row_tiles);
   hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0,
                      stream.stream(), (unsigned*)rowthr.data_ptr<int>(), B);
+  torch::Tensor slab;
+  if (use8p) {
+    slab = torch::empty({(long)nchunks * row_tiles * BM8 * BN8},
+                        torch::TensorOptions().dtype(torch::kFloat32).device(queries.device()));
+  }
   if (k == 1 && !use8p) {
     // assignment fast path: per-row argmax epilogue (no lists/extraction)
     hipLaunchKernelGGL((cosine_topk_partial_t<4>), grid, dim3(THREADS), 0, stream.stream(),
@@ -72,6 +77,7 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
+    (void)slab;
   } else {
     hipLaunchKernelGGL(cosine_topk_partial, grid, dim3(THREADS), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),

@@ -529,12 +529,19 @@ constexpr int BN8 = 256;
 constexpr int THREADS8 = 512;
 constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 
-template <int EPI_MODE>  // 0 = full, 1 = GEMM only
+
+// EPI_MODE: 0 = stash+drain epilogue, 1 = GEMM only, 5 = slab-deferred
+// epilogue (acc quadrants stream to a per-block global slab at tile end;
+// the NEXT tile's windows drain one 32-col slice each, so the top-k
+// maintenance hides under the MFMA pipeline instead of being exposed
+// after the last window).
+template <int EPI_MODE>
 __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
     float* __restrict__ partial_score, int* __restrict__ partial_idx,
     int B, int N, int D, int chunk_tiles, int nchunks,
-    unsigned* rowthr = nullptr, unsigned long long* stats = nullptr) {
+    unsigned* rowthr = nullptr, unsigned long long* stats = nullptr,
+    float* __restrict__ slab = nullptr) {
   __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4];
   char* const smem0 = smem;
   // buffer b in {0,1}: A half h at b*4*HALF8 + h*HALF8; B half h at +2*HALF8
@@ -636,8 +643,69 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 
   const int total_windows = tiles_here * nkt;
 
+  // slab-deferred epilogue state (EPI_MODE 5): this block's 256 KiB slab.
+  // Thread tid<256 owns list row tid; its warm threshold is cached in a
+  // register per tile. The drain is array-free (slot 0 of a list always
+  // holds its min, so the running threshold is one LDS read) to keep the
+  // in-window register footprint tiny next to the 128-reg accumulator.
+  float* myslab = nullptr;
+  if constexpr (EPI_MODE == 5) {
+    const long bid_flat = blockIdx.x + (long)gridDim.x * blockIdx.y;
+    myslab = slab + bid_flat * (BM8 * BN8);
+  }
+  auto drain_slice = [&](int prev_col0, int w, float warm_r) {
+    if (tid >= BM8) return;
+    const int row = tid;
+    volatile float* lrow = lsc + row * KMAX;
+    volatile int* irow = lix + row * KMAX;
+    float rmin = fmaxf(lrow[0], warm_r);  // slot 0 is the list min
+    const float* srow = myslab + row * BN8 + w * 32;
+    bool dirty = false;
+    float mn_out = 0.f;
+#pragma unroll 2
+    for (int ii = 0; ii < 8; ++ii) {
+      const float4 v4 = *(const float4*)(srow + 4 * ii);
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const float v = e == 0 ? v4.x : e == 1 ? v4.y : e == 2 ? v4.z : v4.w;
+        const int gc = prev_col0 + w * 32 + 4 * ii + e;
+        if (v > rmin && gc < N) {
+          // replace slot 0 (the min) with v, then restore the invariant
+          float mn2 = 1e38f;
+          int m2 = 0;
+#pragma unroll
+          for (int q = 1; q < KMAX; ++q) {
+            const float s = lrow[q];
+            if (s < mn2) { mn2 = s; m2 = q; }
+          }
+          if (v <= mn2) {
+            lrow[0] = v;
+            irow[0] = gc;
+          } else {
+            lrow[0] = mn2;
+            irow[0] = irow[m2];
+            lrow[m2] = v;
+            irow[m2] = gc;
+          }
+          const float nmn = fminf(mn2, v);
+          rmin = fmaxf(rmin, nmn);
+          mn_out = nmn;
+          dirty = true;
+        }
+      }
+    }
+    if (dirty && rowthr != nullptr && mn_out > NEG_INF && row0 + row < B)
+      atomicMax(&rowthr[row0 + row], enc_f32(mn_out));
+  };
+
   for (int j = 0; j < tiles_here; ++j) {
     const int col0 = (tile0 + j) * BN8;
+
+    float warm5 = NEG_INF;
+    if constexpr (EPI_MODE == 5) {
+      if (rowthr != nullptr && tid < BM8 && row0 + tid < B)
+        warm5 = dec_f32(rowthr[row0 + tid]);
+    }
 
     f32x4 acc[8][4];
 #pragma unroll
@@ -705,10 +773,29 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         __builtin_amdgcn_s_setprio(0);
         __builtin_amdgcn_s_barrier();  // read-retirement vs next glds
       }
+      if constexpr (EPI_MODE == 5) {
+        // drain one slice of the previous tile's slab under this window's
+        // pipeline (slab stores were separated by >= one window of
+        // barriers; same workgroup -> L1-visible)
+        if (j > 0 && kt < 8) drain_slice(col0 - BN8, kt, warm5);
+      }
     }
 
-    // ---- top-k epilogue: shared lists, serialised by col-quad -----------
-    if constexpr (EPI_MODE == 1) {
+    // ---- top-k epilogue ---------------------------------------------------
+    if constexpr (EPI_MODE == 5) {
+      // stream the final accumulators to the block's slab; the next tile's
+      // windows (or the post-loop flush) merge them into the lists
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg)
+            myslab[(wr * 128 + m * 16 + g * 4 + reg) * BN8 + wc * 64 +
+                   n * 16 + cl] = acc[m][n][reg];
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // slab stores done
+      __builtin_amdgcn_s_barrier();
+    } else if constexpr (EPI_MODE == 1) {
 #pragma unroll
       for (int m = 0; m < 8; ++m)
 #pragma unroll
@@ -817,6 +904,16 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();  // lists settled; stash reusable
       }
+    }
+  }
+
+  if constexpr (EPI_MODE == 5) {
+    if (tiles_here > 0) {
+      const int last_col0 = (tile0 + tiles_here - 1) * BN8;
+      float warmf = NEG_INF;
+      if (rowthr != nullptr && tid < BM8 && row0 + tid < B)
+        warmf = dec_f32(rowthr[row0 + tid]);
+      for (int w = 0; w < 8; ++w) drain_slice(last_col0, w, warmf);
     }
   }
 

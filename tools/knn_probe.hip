@@ -73,8 +73,10 @@ int main(int argc, char** argv) {
   const int nchunks2 = ((ntiles2 + chunk_tiles2 - 1) / chunk_tiles2 + 7) & ~7;
   float* pscore2;
   int* pidx2;
+  float* slab2;
   HIP_CHECK(hipMalloc(&pscore2, (size_t)B * nchunks2 * KMAX * 4));
   HIP_CHECK(hipMalloc(&pidx2, (size_t)B * nchunks2 * KMAX * 4));
+  HIP_CHECK(hipMalloc(&slab2, (size_t)nchunks2 * row_tiles2 * BM8 * BN8 * 4));
   dim3 grid2(nchunks2, row_tiles2);
   auto run_mode = [&](int mode) {
     hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0, 0,
@@ -101,6 +103,9 @@ int main(int argc, char** argv) {
       hipLaunchKernelGGL((cosine_topk_partial8p_t<2>), grid2, dim3(THREADS8), 0, 0,
                          Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
     else if (mode == 4)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<5>), grid2, dim3(THREADS8), 0, 0,
+                         Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr, slab2);
+    else if (mode == 3)
       hipLaunchKernelGGL((cosine_topk_partial8p_t<0>), grid2, dim3(THREADS8), 0, 0,
                          Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
     else
@@ -115,7 +120,7 @@ int main(int argc, char** argv) {
   for (int mi = 0; mi < 5; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[10] = {"full128", "gemm128", "precheck128", "-", "full8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p"};
+  const char* names[10] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p"};
   const int modes[5] = {0, 4, 9, 5, 1};
   std::vector<std::vector<float>> ms(10);
   hipEvent_t t0, t1;
