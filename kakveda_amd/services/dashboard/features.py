@@ -31,16 +31,7 @@ from kakveda_amd.services.dashboard.context import (
     sha256_hex,
 )
 from kakveda_amd.services.dashboard.templates import render
-
-
-def _percentiles(vals: List[float]) -> Dict[str, float]:
-    if not vals:
-        return {"p50": 0.0, "p95": 0.0}
-    vs = sorted(vals)
-    return {
-        "p50": vs[len(vs) // 2],
-        "p95": vs[min(len(vs) - 1, int(len(vs) * 0.95))],
-    }
+from kakveda_amd.utils import percentiles as _percentiles
 
 
 def parse_run_query(q: str) -> Dict[str, Any]:

@@ -172,3 +172,49 @@ async def test_warning_threshold_hot_reload(tmp_path):
     w2 = await cluster.warn("a", prompt)
     assert w2["references"] == []
     await cluster.aclose()
+
+
+async def test_event_bus_drop_on_error(tmp_path):
+    """Fan-out is best-effort: an unreachable subscriber is dropped without
+    failing the publish (reference event_bus/app.py:48-51)."""
+    import httpx
+
+    from kakveda_amd.services import event_bus as eb
+    from kakveda_amd.services.wiring import Transport
+
+    tx = Transport(timeout=0.2)
+    app = eb.create_app(transport=tx)
+    client = httpx.AsyncClient(
+        transport=httpx.ASGITransport(app=app), base_url="http://bus"
+    )
+    await client.post(
+        "/subscribe",
+        json={"topic": "t", "callback_url": "http://127.0.0.1:59999/nope"},
+    )
+    resp = await client.post("/publish", json={"topic": "t", "payload": {"x": 1}})
+    body = resp.json()
+    assert body["ok"] and body["subscribers"] == 1 and body["delivered"] == 0
+    # duplicate subscription dedups
+    await client.post(
+        "/subscribe",
+        json={"topic": "t", "callback_url": "http://127.0.0.1:59999/nope"},
+    )
+    topics = (await client.get("/topics")).json()["topics"]
+    assert len(topics["t"]) == 1
+    await client.aclose()
+    await tx.aclose()
+
+
+def test_stopwatch_and_percentiles():
+    from kakveda_amd.utils import Stopwatch, percentiles
+
+    sw = Stopwatch()
+    with sw.span("a"):
+        pass
+    with sw.span("b"):
+        pass
+    assert [s["name"] for s in sw.spans] == ["a", "b"]
+    assert sw.total_ms() >= 0
+    p = percentiles([3.0, 1.0, 2.0, 4.0])
+    assert p["p50"] == 3.0 and p["p95"] == 4.0
+    assert percentiles([]) == {"p50": 0.0, "p95": 0.0}
