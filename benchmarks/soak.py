@@ -1,0 +1,119 @@
+"""Sustained-serving soak: concurrent inserts + matches against a live
+GFKB engine on GPU (exercises the search-while-insert path: engine lock +
+valid_n prefix scanning), followed by an exactness audit.
+
+Run: python benchmarks/soak.py [--seconds 60] [--entries 2000000]
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import threading
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seconds", type=float, default=60.0)
+    ap.add_argument("--entries", type=int, default=2_000_000)
+    ap.add_argument("--batch", type=int, default=1024)
+    ap.add_argument("--dim", type=int, default=768)
+    args = ap.parse_args()
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    from kakveda_amd import ops
+    from kakveda_amd.gfkb.engine import EmbeddingStore
+
+    store = EmbeddingStore(args.dim, device=device, capacity=args.entries * 2)
+    gen = torch.Generator(device=device).manual_seed(77)
+    base = torch.randn(args.entries, args.dim, generator=gen, device=device).to(
+        store.dtype
+    )
+    if device == "cuda":
+        ops.l2normalize_(base)
+    store.append(base)
+
+    lock = threading.Lock()
+    stop = threading.Event()
+    inserted = [0]
+    errors: list[str] = []
+
+    def inserter():
+        g2 = torch.Generator(device=device).manual_seed(99)
+        while not stop.is_set():
+            rows = torch.randn(256, args.dim, generator=g2, device=device).to(store.dtype)
+            if device == "cuda":
+                ops.l2normalize_(rows)
+            with lock:
+                store.append(rows)
+                inserted[0] += 256
+            time.sleep(0.01)
+
+    t = threading.Thread(target=inserter, daemon=True)
+    t.start()
+
+    lookups = 0
+    lat = []
+    qgen = torch.Generator(device=device).manual_seed(5)
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < args.seconds:
+        q = torch.randn(args.batch, args.dim, generator=qgen, device=device).to(store.dtype)
+        if device == "cuda":
+            ops.l2normalize_(q)
+        ts = time.perf_counter()
+        with lock:
+            scores, idx = store.search(q, 5)
+            n_seen = store.count
+        if device == "cuda":
+            torch.cuda.synchronize()
+        lat.append(time.perf_counter() - ts)
+        lookups += args.batch
+        # light invariants every step
+        if bool((idx >= n_seen).any()) or bool((idx < 0).any()):
+            errors.append(f"idx out of live prefix at count={n_seen}")
+            break
+        if bool(torch.isinf(scores).any() and (scores > 1.5).any()):
+            errors.append("implausible score")
+            break
+    stop.set()
+    t.join(timeout=5)
+    elapsed = time.perf_counter() - t0
+
+    # final exactness audit vs torch on the final store state
+    q = torch.randn(64, args.dim, generator=qgen, device=device).to(store.dtype)
+    if device == "cuda":
+        ops.l2normalize_(q)
+    scores, idx = store.search(q, 5)
+    sims = q.float() @ store.data[: store.count].float().t()
+    ref, _ = torch.topk(sims, 5, dim=1)
+    exact = bool(torch.allclose(scores, ref, atol=2e-2, rtol=1e-2))
+
+    lat_ms = sorted(x * 1000 for x in lat)
+    print(
+        json.dumps(
+            {
+                "metric": "soak",
+                "seconds": round(elapsed, 1),
+                "lookups": lookups,
+                "lookups_per_sec": round(lookups / elapsed),
+                "inserted_rows": inserted[0],
+                "final_store_rows": store.count,
+                "p50_ms": round(lat_ms[len(lat_ms) // 2], 2),
+                "p99_ms": round(lat_ms[max(0, int(len(lat_ms) * 0.99) - 1)], 2),
+                "errors": errors,
+                "final_exactness_vs_torch": exact,
+            }
+        )
+    )
+    return 1 if errors or not exact else 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
