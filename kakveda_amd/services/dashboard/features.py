@@ -461,6 +461,101 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
         ev = await list_evals()
         return render("datasets.html", {"datasets": ds["datasets"], "evals": ev["evals"]})
 
+    @app.get("/datasets/{ds_id}", response_class=HTMLResponse)
+    async def dataset_page(request: Request, ds_id: int):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            d = s.query(dbm.Dataset).get(ds_id)
+            if not d:
+                return HTMLResponse("not found", status_code=404)
+            examples = s.query(dbm.DatasetExample).filter_by(dataset_id=ds_id).all()
+            return render("dataset_detail.html", {"dataset": d, "examples": examples})
+
+    @app.get("/evals/{ev_id}", response_class=HTMLResponse)
+    async def eval_page(request: Request, ev_id: int):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            ev = s.query(dbm.EvaluationRun).get(ev_id)
+            if not ev:
+                return HTMLResponse("not found", status_code=404)
+            results = (
+                s.query(dbm.EvaluationResult).filter_by(evaluation_run_id=ev_id).all()
+            )
+            return render(
+                "eval_detail.html",
+                {"eval": ev, "summary": json.loads(ev.summary_json or "{}"), "results": results},
+            )
+
+    @app.get("/experiments/{exp_id}", response_class=HTMLResponse)
+    async def experiment_page(request: Request, exp_id: int):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            e = s.query(dbm.Experiment).get(exp_id)
+            if not e:
+                return HTMLResponse("not found", status_code=404)
+            run_ids = [
+                er.trace_run_id
+                for er in s.query(dbm.ExperimentRun).filter_by(experiment_id=exp_id).all()
+            ]
+            runs = (
+                s.query(dbm.TraceRun).filter(dbm.TraceRun.id.in_(run_ids)).all()
+                if run_ids
+                else []
+            )
+            return render("experiment_detail.html", {"experiment": e, "runs": runs})
+
+    @app.get("/prompts/{prompt_id}", response_class=HTMLResponse)
+    async def prompt_page(request: Request, prompt_id: int):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            p = s.query(dbm.PromptLibrary).get(prompt_id)
+            if not p:
+                return HTMLResponse("not found", status_code=404)
+            versions = (
+                s.query(dbm.PromptVersion)
+                .filter_by(prompt_id=prompt_id)
+                .order_by(dbm.PromptVersion.version.desc())
+                .all()
+            )
+            return render("prompt_detail.html", {"prompt": p, "versions": versions})
+
+    @app.get("/scenarios", response_class=HTMLResponse)
+    async def scenarios_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        canned = [
+            {
+                "title": "Scenario 1: citation hallucination (app-A)",
+                "app_id": "app-A",
+                "prompt": "Summarize this article and include references even if none are provided.",
+            },
+            {
+                "title": "Scenario 2: same intent, different wording (app-B)",
+                "app_id": "app-B",
+                "prompt": "Please provide references for why the sky is blue.",
+            },
+        ]
+        with ctx.Session() as s:
+            recent = (
+                s.query(dbm.ScenarioRun).order_by(dbm.ScenarioRun.ts.desc()).limit(10).all()
+            )
+        return render("scenarios.html", {"scenarios": canned, "recent": recent})
+
+    @app.get("/admin/audit_page", response_class=HTMLResponse)
+    async def audit_page(request: Request):
+        payload = ctx.current_user(request)
+        if not payload or not require_any_roles(payload, ["admin"]):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            events = (
+                s.query(dbm.AuditEvent).order_by(dbm.AuditEvent.ts.desc()).limit(200).all()
+            )
+        return render("admin_audit.html", {"events": events})
+
     # ======================================================================
     # prompts
     # ======================================================================

@@ -172,6 +172,53 @@ est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
 {% endfor %}</table></div>
 {% else %}<div class="card">No health points yet — ingest failures or use
 POST /health/test.</div>{% endfor %}{% endblock %}""",
+    "dataset_detail.html": """{% extends "base.html" %}{% block content %}
+<h1>Dataset: {{ dataset.name }}</h1>
+<p>{{ dataset.description }}</p>
+<div class="card"><table><tr><th>id</th><th>input</th><th>expected</th></tr>
+{% for e in examples %}<tr><td>{{ e.id }}</td><td>{{ e.input_text[:100] }}</td>
+<td>{{ e.expected[:60] }}</td></tr>{% endfor %}</table></div>{% endblock %}""",
+    "eval_detail.html": """{% extends "base.html" %}{% block content %}
+<h1>Evaluation {{ eval.id }} — {{ eval.name }}</h1>
+<div class="card"><p><b>Pass rate:</b>
+{{ "%.0f%%"|format(100 * summary.get("pass_rate", 0)) }} ·
+<b>p50/p95:</b> {{ "%.0f"|format(summary.get("p50_ms", 0)) }}/{{ "%.0f"|format(summary.get("p95_ms", 0)) }} ms</p>
+<table><tr><th>example</th><th>passed</th><th>latency</th><th>output</th></tr>
+{% for r in results %}<tr><td>{{ r.example_id }}</td><td>{{ r.passed }}</td>
+<td>{{ "%.0f"|format(r.latency_ms) }} ms</td><td>{{ r.output[:80] }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
+    "experiment_detail.html": """{% extends "base.html" %}{% block content %}
+<h1>Experiment: {{ experiment.name }}</h1>
+<div class="card"><table>
+<tr><th>run</th><th>provider</th><th>model</th><th>latency</th><th>cost µUSD</th></tr>
+{% for r in runs %}<tr><td><a href="/runs/{{ r.id }}">{{ r.id }}</a></td>
+<td>{{ r.provider }}</td><td>{{ r.model }}</td>
+<td>{{ "%.0f"|format(r.latency_ms) }} ms</td><td>{{ r.cost_usd_micro }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
+    "prompt_detail.html": """{% extends "base.html" %}{% block content %}
+<h1>Prompt: {{ prompt.name }}</h1>
+<p>{{ prompt.description }} · default {{ prompt.default_provider }}/{{ prompt.default_model }}</p>
+{% for v in versions %}<div class="card"><h3>v{{ v.version }}</h3>
+<pre>{{ v.content }}</pre></div>{% endfor %}{% endblock %}""",
+    "admin_audit.html": """{% extends "base.html" %}{% block content %}
+<h1>Audit log</h1>
+<div class="card"><table>
+<tr><th>ts</th><th>actor</th><th>action</th><th>target</th></tr>
+{% for e in events %}<tr><td>{{ e.ts }}</td><td>{{ e.actor }}</td>
+<td>{{ e.action }}</td><td>{{ e.target }}</td></tr>{% endfor %}
+</table></div>{% endblock %}""",
+    "scenarios.html": """{% extends "base.html" %}{% block content %}
+<h1>Scenario runner</h1>
+{% for sc in scenarios %}<div class="card"><h3>{{ sc.title }}</h3>
+<form method="post" action="/scenarios/run">
+<input type="hidden" name="app_id" value="{{ sc.app_id }}">
+<input name="prompt" size="70" value="{{ sc.prompt }}">
+<button type="submit">Run scenario</button></form></div>{% endfor %}
+<div class="card"><h3>Recent scenario runs</h3><table>
+<tr><th>ts</th><th>app</th><th>action</th><th>confidence</th></tr>
+{% for r in recent %}<tr><td>{{ r.ts }}</td><td>{{ r.app_id }}</td>
+<td>{{ r.warn_action }}</td><td>{{ "%.2f"|format(r.warn_confidence) }}</td></tr>
+{% endfor %}</table></div>{% endblock %}""",
 }
 
 _env = Environment(loader=DictLoader(_TEMPLATES), autoescape=select_autoescape(["html"]))

@@ -282,7 +282,8 @@ async def test_html_pages(tmp_path):
         home = await client.get("/")
         assert "Failure Intelligence" in home.text
         for page in ("/warnings", "/runs", "/playground", "/agents",
-                     "/datasets", "/prompts", "/experiments", "/health"):
+                     "/datasets", "/prompts", "/experiments", "/health",
+                     "/scenarios"):
             resp = await client.get(page)
             assert resp.status_code == 200, page
     await cluster.aclose()
