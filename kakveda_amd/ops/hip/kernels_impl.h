@@ -492,38 +492,6 @@ inline constexpr auto cosine_topk_partial = cosine_topk_partial_t<0, 2>;
 // ===========================================================================
 
 
-// Per-m epilogue body for the 8-phase kernel: __noinline__ so the 8 m-bodies
-// don't hoist 128 candidate values live on top of the 128-register
-// accumulator (measured: inlined epilogue = 256 VGPR + 384 B/lane scratch
-// with spill code inside the K-loop, 4x slowdown).
-template <bool PUBLISH, bool DO_EXTRACT = true>
-__device__ __noinline__ void topk_epilogue_m8(
-    volatile float* lsc, volatile int* lix, f32x4 a0, f32x4 a1, f32x4 a2,
-    f32x4 a3, int rowbase, int rlbase, float warm, float warm2, int colb,
-    int N, int lane, int g, unsigned* rowthr, int B, int row0) {
-#pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    const int rl = rlbase + g * 4 + reg;
-    const int row = rowbase + g * 4 + reg;
-    const float rwarm = __shfl(rl < 64 ? warm : warm2, rl & 63, 64);
-    const int lbase = row * KMAX;
-    const float rmin0 = fmaxf(lsc[lbase], rwarm);
-    float w0 = (colb + 0 < N) ? a0[reg] : NEG_INF;
-    float w1 = (colb + 16 < N) ? a1[reg] : NEG_INF;
-    float w2 = (colb + 32 < N) ? a2[reg] : NEG_INF;
-    float w3 = (colb + 48 < N) ? a3[reg] : NEG_INF;
-    float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
-#pragma unroll
-    for (int off = 1; off < 16; off <<= 1)
-      gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
-    if (DO_EXTRACT && gmax > rmin0) {
-      topk_extract_group<PUBLISH>(lsc, lix, lbase, rwarm, w0, w1, w2, w3,
-                                  colb, N, lane, g, rowthr,
-                                  (row0 + row < B) ? row0 + row + 1 : 0);
-    }
-  }
-}
-
 constexpr int BM8 = 256;
 constexpr int BN8 = 256;
 constexpr int THREADS8 = 512;
