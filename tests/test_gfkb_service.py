@@ -99,5 +99,5 @@ async def test_patterns_endpoints(tmp_path):
     assert len(pats) == 1 and sorted(pats[0]["failure_ids"]) == ["F-0001", "F-0002"]
 
     hz = (await client.get("/healthz")).json()
-    assert hz["ok"] and hz["rows"] == 2
+    assert hz["ok"] and hz["rows"] == 0  # patterns only: no embedding rows
     await client.aclose()
