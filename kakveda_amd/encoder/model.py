@@ -14,9 +14,8 @@ the same math in plain torch (deterministic, used by BASELINE config 1).
 
 from __future__ import annotations
 
-from typing import Optional, Sequence
+from typing import Sequence
 
-import numpy as np
 import torch
 
 from kakveda_amd.encoder.featurizer import featurize_batch

@@ -11,7 +11,7 @@ latency-bound; compute-heavy local search dominates.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.distributed as td

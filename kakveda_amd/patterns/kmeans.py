@@ -10,7 +10,7 @@ renormalising (BASELINE config 4).
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.distributed as td

@@ -7,9 +7,7 @@ services/pattern_detector/app.py:29-58 generalised per SURVEY.md 2.5).
 from __future__ import annotations
 
 from collections import defaultdict
-from typing import Dict, List, Optional
-
-import torch
+from typing import Dict, List
 
 from kakveda_amd.gfkb.engine import GfkbEngine
 from kakveda_amd.patterns.kmeans import StreamingKMeans

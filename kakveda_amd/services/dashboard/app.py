@@ -17,7 +17,6 @@ from fastapi import FastAPI, Request
 from fastapi.responses import HTMLResponse, JSONResponse, RedirectResponse, Response
 
 from kakveda_amd.core.schemas import utcnow
-from kakveda_amd.core.signature import detect_citation_markers
 from kakveda_amd.services import (
     TOPIC_CHILD_SAFETY,
     TOPIC_FAILURE_DETECTED,

@@ -13,7 +13,6 @@ from __future__ import annotations
 import datetime as dt
 import json
 import secrets
-import statistics
 import time
 import uuid
 from typing import Any, Dict, List, Optional
