@@ -58,21 +58,37 @@ def _env_for(name: str, data_dir: str) -> dict:
     return env
 
 
+def _ask(prompt_text: str, default: str) -> str:
+    """Interactive question (reference kakveda_cli/prompts.py flow);
+    falls back to the default when stdin is not a TTY or --yes is set."""
+    try:
+        raw = input(f"{prompt_text} [{default}]: ").strip()
+        return raw or default
+    except EOFError:
+        return default
+
+
 def cmd_init(args) -> int:
-    """Interactive-ish .env generation (reference kakveda_cli/config.py)."""
+    """.env generation; interactive when on a TTY (reference
+    kakveda_cli/prompts.py + config.py), flag-driven otherwise."""
     path = Path(".env")
     if path.exists() and not args.force:
         print(".env exists; use --force to overwrite")
         return 1
+    env_name, data_dir, model_url = args.env, args.data_dir, ""
+    if sys.stdin.isatty() and not args.yes:
+        env_name = _ask("environment (dev/prod)", env_name)
+        data_dir = _ask("data directory", data_dir)
+        model_url = _ask("Ollama URL (empty = deterministic stub)", "")
     lines = [
-        f"KAKVEDA_ENV={args.env}",
+        f"KAKVEDA_ENV={env_name}",
         f"KAKVEDA_JWT_SECRET={os.urandom(24).hex()}",
-        f"DATA_DIR={args.data_dir}",
+        f"DATA_DIR={data_dir}",
         "KAKVEDA_LOG_FORMAT=json",
-        "# OLLAMA_URL=http://127.0.0.1:11434",
+        f"OLLAMA_URL={model_url}" if model_url else "# OLLAMA_URL=http://127.0.0.1:11434",
     ]
     path.write_text("\n".join(lines) + "\n")
-    print(f"wrote {path} ({args.env})")
+    print(f"wrote {path} ({env_name})")
     return 0
 
 
@@ -205,10 +221,11 @@ def main(argv=None) -> int:
     ap = argparse.ArgumentParser(prog="kakveda-amd")
     sub = ap.add_subparsers(dest="cmd", required=True)
 
-    p = sub.add_parser("init", help="write a .env")
+    p = sub.add_parser("init", help="write a .env (interactive on a TTY)")
     p.add_argument("--env", default="dev", choices=["dev", "prod"])
     p.add_argument("--data-dir", default="./data")
     p.add_argument("--force", action="store_true")
+    p.add_argument("--yes", action="store_true", help="accept defaults, no prompts")
     p.set_defaults(fn=cmd_init)
 
     p = sub.add_parser("up", help="start all services as local processes")

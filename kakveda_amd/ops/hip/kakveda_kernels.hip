@@ -40,7 +40,10 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   const int row_tiles = (B + tile_m - 1) / tile_m;
   const int ntiles = (N + tile_n - 1) / tile_n;
   // size chunks so the grid comfortably oversubscribes 256 CUs
-  const long target = use8p ? 512 : 2048;
+  // (KAKVEDA_KNN_TARGET overrides the target block count for tuning)
+  static const char* tenv = std::getenv("KAKVEDA_KNN_TARGET");
+  static const long tover = tenv ? std::atol(tenv) : 0;
+  const long target = tover > 0 ? tover : (use8p ? 512 : 2048);
   long want = ((long)ntiles * row_tiles + target - 1) / target;
   const int chunk_tiles = (int)std::max(4L, std::min(want, 128L));
   // pad the chunk count to a multiple of 8 so the in-kernel XCD remap is
