@@ -66,11 +66,13 @@ def main() -> None:
 
     import uvicorn
 
+    from kakveda_amd.core.metrics import instrument as instrument_metrics
     from kakveda_amd.core.otel import instrument_fastapi, setup_otel
 
     app = build_app(args.service)
     setup_otel(f"kakveda-{args.service.replace('_', '-')}")
     instrument_fastapi(app)
+    instrument_metrics(app, args.service)
     uvicorn.run(
         app,
         host=args.host,

@@ -61,9 +61,11 @@ def create_app(
 
     @app.post("/failures/match", response_model=FailureMatchResponse)
     async def match(req: FailureMatchRequest):
-        return FailureMatchResponse(
-            matches=engine.match(req.signature_text, failure_type=req.failure_type)
-        )
+        from kakveda_amd.core.metrics import observe_gfkb
+
+        matches = engine.match(req.signature_text, failure_type=req.failure_type)
+        observe_gfkb(engine.store.count, len(engine.failures))
+        return FailureMatchResponse(matches=matches)
 
     @app.post("/failures/upsert")
     async def upsert(req: UpsertFailureRequest):
@@ -94,6 +96,9 @@ def create_app(
 
     @app.get("/healthz")
     async def healthz():
+        from kakveda_amd.core.metrics import observe_gfkb
+
+        observe_gfkb(engine.store.count, len(engine.failures))
         return {"ok": True, "failures": len(engine.failures), "rows": engine.store.count}
 
     return app

@@ -58,7 +58,10 @@ def create_app(
         except Exception:
             pattern_id = None
 
+        from kakveda_amd.core.metrics import observe_warn
+
         if best and score >= threshold:
+            observe_warn(action_default)
             msg = (
                 f"This execution matches past failure type {best.get('failure_type')} "
                 f"(failure_id={best.get('failure_id')}, similarity={score:.2f}). "
@@ -71,6 +74,7 @@ def create_app(
                 references=[best],
                 message=msg,
             )
+        observe_warn("silent" if action_default == "silent" else "warn")
         return WarningResponse(
             action="silent" if action_default == "silent" else "warn",
             confidence=score,

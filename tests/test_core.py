@@ -218,3 +218,24 @@ def test_stopwatch_and_percentiles():
     p = percentiles([3.0, 1.0, 2.0, 4.0])
     assert p["p50"] == 3.0 and p["p95"] == 4.0
     assert percentiles([]) == {"p50": 0.0, "p95": 0.0}
+
+
+async def test_prometheus_metrics_endpoint(tmp_path):
+    """Every served app exposes /metrics with request counters
+    (exceeds reference parity — SURVEY.md 5.5 notes 'No Prometheus')."""
+    import httpx
+
+    from kakveda_amd.core.metrics import instrument
+    from kakveda_amd.services import event_bus as eb
+
+    app = eb.create_app()
+    assert instrument(app, "event-bus-test")
+    client = httpx.AsyncClient(
+        transport=httpx.ASGITransport(app=app), base_url="http://x"
+    )
+    await client.get("/topics")
+    metrics = (await client.get("/metrics")).text
+    assert "kakveda_http_requests_total" in metrics
+    assert 'service="event-bus-test"' in metrics
+    assert "kakveda_http_request_seconds" in metrics
+    await client.aclose()
