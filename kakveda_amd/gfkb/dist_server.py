@@ -1,0 +1,158 @@
+"""Distributed GFKB serving: one process per GPU, rank 0 fronts HTTP.
+
+The bench proves the sharded data plane; this module makes it a *service*
+(SURVEY.md 2.5 'sharded GFKB'): rank 0 runs the FastAPI app and broadcasts
+every store-mutating or collective operation as a command over
+torch.distributed (RCCL on GPUs, gloo on CPU); ranks 1..N-1 sit in a
+worker loop executing the same ShardedStore calls SPMD-style, so the
+all-gather merge inside ``ShardedStore.search`` lines up across ranks.
+
+Launch (8 GPUs):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \\
+      --master-addr 127.0.0.1 -m kakveda_amd.gfkb.dist_server --port 8101
+
+Commands are tiny control tuples; payloads (embedding batches) travel as
+broadcast tensors.
+"""
+
+from __future__ import annotations
+
+import argparse
+from typing import Optional, Tuple
+
+import torch
+import torch.distributed as td
+
+from kakveda_amd.parallel.dist import init_from_env
+from kakveda_amd.parallel.sharded import ShardedStore
+
+CMD_APPEND = "append"
+CMD_SEARCH = "search"
+CMD_COUNT = "count"
+CMD_STOP = "stop"
+
+
+class DistGfkbCoordinator:
+    """Rank-0 handle: broadcasts a command, then performs the same local
+    collective call every worker performs."""
+
+    def __init__(self, dim: int = 768, capacity: int = 1 << 20):
+        self.rank, self.world, self.device = init_from_env()
+        self.dim = dim
+        self.store = ShardedStore(dim, device=str(self.device), capacity=capacity)
+
+    # -- plumbing ----------------------------------------------------------
+
+    def _bcast_cmd(self, cmd: str, meta: Tuple = ()) -> None:
+        if self.world > 1:
+            obj = [cmd, meta]
+            td.broadcast_object_list(obj, src=0)
+
+    def _bcast_tensor(self, t: torch.Tensor) -> torch.Tensor:
+        if self.world > 1:
+            td.broadcast(t, src=0)
+        return t
+
+    # -- operations (call on rank 0 only) ----------------------------------
+
+    def append(self, rows: torch.Tensor) -> int:
+        rows = rows.to(self.device)
+        self._bcast_cmd(CMD_APPEND, (rows.shape[0], rows.shape[1], str(rows.dtype)))
+        self._bcast_tensor(rows.contiguous())
+        return self.store.append(rows)
+
+    def search(self, queries: torch.Tensor, k: int):
+        q = queries.to(self.device)
+        self._bcast_cmd(CMD_SEARCH, (q.shape[0], q.shape[1], int(k), str(q.dtype)))
+        self._bcast_tensor(q.contiguous())
+        return self.store.search(q, k)
+
+    def total(self) -> int:
+        return self.store.total
+
+    def stop(self) -> None:
+        self._bcast_cmd(CMD_STOP)
+
+
+def worker_loop(coord_dim: int = 768, capacity: int = 1 << 20) -> None:
+    """Ranks 1..N-1: execute broadcast commands until CMD_STOP."""
+    rank, world, device = init_from_env()
+    store = ShardedStore(coord_dim, device=str(device), capacity=capacity)
+
+    def recv_tensor(n: int, d: int, dtype_str: str) -> torch.Tensor:
+        dtype = getattr(torch, dtype_str.replace("torch.", ""))
+        t = torch.empty(n, d, dtype=dtype, device=device)
+        td.broadcast(t, src=0)
+        return t
+
+    while True:
+        obj = [None, None]
+        td.broadcast_object_list(obj, src=0)
+        cmd, meta = obj
+        if cmd == CMD_STOP:
+            break
+        if cmd == CMD_APPEND:
+            n, d, dt = meta
+            store.append(recv_tensor(n, d, dt))
+        elif cmd == CMD_SEARCH:
+            n, d, k, dt = meta
+            store.search(recv_tensor(n, d, dt), k)
+        elif cmd == CMD_COUNT:
+            pass
+
+
+def main(argv: Optional[list] = None) -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--port", type=int, default=8101)
+    ap.add_argument("--dim", type=int, default=768)
+    ap.add_argument("--data-dir", default="./data")
+    args = ap.parse_args(argv)
+
+    rank, world, device = init_from_env()
+    if rank != 0:
+        worker_loop(args.dim)
+        return
+
+    # rank 0: the HTTP service on top of a distributed engine. The engine's
+    # EmbeddingStore is replaced by the coordinator-backed sharded store.
+    import uvicorn
+
+    from kakveda_amd.gfkb.engine import GfkbEngine
+    from kakveda_amd.services.gfkb_service import create_app
+
+    coord = DistGfkbCoordinator(dim=args.dim)
+    engine = GfkbEngine(data_dir=args.data_dir, device=str(device), dim=args.dim)
+    engine.store = _CoordinatorStore(coord)  # type: ignore[assignment]
+    app = create_app(engine=engine)
+    try:
+        uvicorn.run(app, host="0.0.0.0", port=args.port)
+    finally:
+        coord.stop()
+
+
+class _CoordinatorStore:
+    """EmbeddingStore facade over the coordinator (duck-typed)."""
+
+    def __init__(self, coord: DistGfkbCoordinator):
+        self.coord = coord
+        self.dim = coord.dim
+        self.device = coord.device
+        self.dtype = coord.store.local.dtype
+
+    @property
+    def count(self) -> int:
+        return self.coord.total()
+
+    def append(self, rows: torch.Tensor) -> int:
+        return self.coord.append(rows.to(self.dtype))
+
+    def search(self, queries: torch.Tensor, k: int):
+        return self.coord.search(queries.to(self.dtype), k)
+
+    @property
+    def data(self) -> torch.Tensor:
+        return self.coord.store.local.data
+
+
+if __name__ == "__main__":
+    main()
