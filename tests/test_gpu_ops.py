@@ -204,3 +204,33 @@ def test_kmeans_update_kernel_vs_torch():
     assert torch.allclose(sums, ref_s, atol=0.5, rtol=1e-2), (
         (sums - ref_s).abs().max().item()
     )
+
+
+def test_cosine_topk_argument_validation():
+    from kakveda_amd import ops
+
+    q = _rand_unit(8, 768, seed=40)
+    c = _rand_unit(100, 768, seed=41)
+    with pytest.raises(RuntimeError):
+        ops.cosine_topk(q, c, 9)  # k > KMAX
+    with pytest.raises(RuntimeError):
+        ops.cosine_topk(q.float(), c, 5)  # wrong dtype
+    q100 = torch.randn(8, 100, device=_dev(), dtype=torch.bfloat16)
+    c100 = torch.randn(50, 100, device=_dev(), dtype=torch.bfloat16)
+    with pytest.raises(RuntimeError):
+        ops.cosine_topk(q100, c100, 5)  # D not a multiple of 64
+
+
+def test_k1_argmax_fast_path():
+    """k=1 takes the dedicated argmax epilogue; results must match torch."""
+    from kakveda_amd import ops
+
+    q = _rand_unit(512, 768, seed=42)
+    c = _rand_unit(50000, 768, seed=43)
+    scores, idx = ops.cosine_topk(q, c, 1)
+    torch.cuda.synchronize()
+    sims = q.float() @ c.float().t()
+    ref_s, ref_i = sims.max(dim=1)
+    assert torch.allclose(scores[:, 0], ref_s, atol=2e-2, rtol=1e-2)
+    gathered = sims.gather(1, idx)
+    assert torch.allclose(gathered[:, 0], scores[:, 0], atol=1e-4)
