@@ -122,7 +122,9 @@ def main(argv: Optional[list] = None) -> None:
 
     coord = DistGfkbCoordinator(dim=args.dim)
     engine = GfkbEngine(data_dir=args.data_dir, device=str(device), dim=args.dim)
-    engine.store = _CoordinatorStore(coord)  # type: ignore[assignment]
+    # migrate JSONL-restored rows into the coordinator-backed store (the
+    # engine's __init__ rebuilt them into the throwaway local store)
+    engine.attach_store(_CoordinatorStore(coord))  # type: ignore[arg-type]
     app = create_app(engine=engine)
     try:
         uvicorn.run(app, host="0.0.0.0", port=args.port)

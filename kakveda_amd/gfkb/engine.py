@@ -131,6 +131,18 @@ class GfkbEngine:
             self._row_identity = identities
             self._identity_row = {key: i for i, key in enumerate(identities)}
 
+    def attach_store(self, new_store) -> None:
+        """Swap the embedding store (e.g. for a distributed coordinator
+        store) and re-encode every known identity into it, preserving row
+        order. Must happen before serving begins."""
+        with self._lock:
+            self.store = new_store
+            if self._row_identity:
+                emb = self.encoder.encode_texts(
+                    [sig for (_ft, sig) in self._row_identity]
+                )
+                self.store.append(emb)
+
     # -- failures ----------------------------------------------------------
 
     def list_failures(self) -> List[Dict[str, Any]]:

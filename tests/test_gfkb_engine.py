@@ -96,3 +96,20 @@ def test_store_growth(tmp_path):
         "intent_tags: | prompt_hint:unique prompt number 17 | tools: | env_keys:"
     )
     assert m[0].score >= 0.99
+
+
+def test_attach_store_migrates_rows(tmp_path):
+    """Swapping the embedding store re-encodes known identities (the
+    distributed server path: JSONL-restored rows must survive)."""
+    from kakveda_amd.gfkb.engine import EmbeddingStore
+
+    eng = _engine(tmp_path)
+    eng.upsert_failure("HALLUCINATION_CITATION", SIG, {}, app_id="a")
+    eng.upsert_failure(
+        "T", "intent_tags: | prompt_hint:other | tools: | env_keys:", {}, app_id="b"
+    )
+    fresh = EmbeddingStore(256, device="cpu", capacity=64)
+    eng.attach_store(fresh)
+    assert eng.store is fresh and eng.store.count == 2
+    m = eng.match(SIG)
+    assert m and m[0].failure_id == "F-0001" and m[0].score >= 0.99
