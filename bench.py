@@ -50,18 +50,25 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    on_gpu = torch.cuda.is_available()
     dist = world > 1
     if dist:
         import torch.distributed as td
 
-        td.init_process_group(backend="nccl")
-    device = torch.device(f"cuda:{local_rank}")
-    torch.cuda.set_device(device)
+        td.init_process_group(backend="nccl" if on_gpu else "gloo")
+    if on_gpu:
+        device = torch.device(f"cuda:{local_rank}")
+        torch.cuda.set_device(device)
+    else:
+        # CPU dry-run mode: same code path (sharding, all-gather merge)
+        # over gloo with the torch fallback ops — used to validate the
+        # exact torchrun launch contract without a GPU.
+        device = torch.device("cpu")
 
     from kakveda_amd import ops
     from kakveda_amd.parallel.sharded import ShardedStore
 
-    if not ops.hip_available():
+    if on_gpu and not ops.hip_available():
         raise RuntimeError("HIP extension not built; run __graft_entry__.build() first")
 
     D, B, k = args.dim, args.batch, args.topk
@@ -69,12 +76,13 @@ def main():
 
     # ---- build the sharded GFKB (synthetic fingerprints, bf16 in HBM) ----
     gen = torch.Generator(device=device).manual_seed(1000 + rank)
-    corpus = torch.empty(per_rank, D, dtype=torch.bfloat16, device=device)
+    cdtype = torch.bfloat16 if on_gpu else torch.float32
+    corpus = torch.empty(per_rank, D, dtype=cdtype, device=device)
     fill = 1 << 20
     for s in range(0, per_rank, fill):
         e = min(s + fill, per_rank)
         corpus[s:e] = torch.randn(e - s, D, generator=gen, device=device, dtype=torch.float32).to(
-            torch.bfloat16
+            cdtype
         )
     ops.l2normalize_(corpus)
     store = ShardedStore(D, device=str(device), capacity=1024)
@@ -99,7 +107,7 @@ def main():
 
     def step() -> int:
         # 1. encode on GPU (embedding_bag kernel + projection GEMM + norm)
-        q = enc.encode_features(feat_idx, feat_w).to(torch.bfloat16)
+        q = enc.encode_features(feat_idx, feat_w).to(cdtype)
         # 2-4. fused cosine top-k per shard, RCCL all-gather over xGMI,
         #      exact (score, global-id) merge — kakveda_amd.parallel
         fs, fid = store.search(q, k)
@@ -115,7 +123,8 @@ def main():
         import torch.distributed as td
 
         td.barrier()
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
 
     # ---- timed region ----------------------------------------------------
     times = []
@@ -123,13 +132,15 @@ def main():
     for _ in range(args.steps):
         ts = time.perf_counter()
         step()
-        torch.cuda.synchronize()
+        if on_gpu:
+            torch.cuda.synchronize()
         times.append(time.perf_counter() - ts)
     if dist:
         import torch.distributed as td
 
         td.barrier()
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
     if dist:
@@ -155,7 +166,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if on_gpu else "fp32-cpu-dryrun",
             "data": "synthetic",
             "config": {
                 "model": "gfkb-cosine-knn-768d",

@@ -781,7 +781,7 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
             row = s.query(dbm.ProjectApiKey).filter_by(key_hash=h, revoked=False).first()
             if not row:
                 return None
-            return s.query(dbm.Project).get(row.project_id)
+            return s.get(dbm.Project, row.project_id)
 
     @app.post("/api/agents/register")
     async def api_register_agent(request: Request):
@@ -859,7 +859,7 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
             return JSONResponse({"ok": False}, status_code=403)
         body = await read_payload(request)
         with ctx.Session() as s:
-            if not s.query(dbm.Project).get(proj_id):
+            if not s.get(dbm.Project, proj_id):
                 return JSONResponse({"ok": False}, status_code=404)
             plaintext = "kv-" + secrets.token_urlsafe(24)
             s.add(
