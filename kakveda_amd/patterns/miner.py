@@ -22,13 +22,20 @@ class PatternMiner:
     def mine(self, iters: int = 8) -> List[Dict]:
         """Cluster the live GFKB rows; upsert a pattern per cluster whose
         members span >= min_apps apps. Returns the upserted patterns."""
-        store = self.engine.store
-        n = store.count
-        if n < 2:
-            return []
+        # snapshot rows + their latest records under the engine lock so a
+        # concurrent upsert can't shift row identities mid-clustering;
+        # the (GPU) k-means itself runs outside the lock.
+        with self.engine._lock:
+            store = self.engine.store
+            n = store.count
+            if n < 2:
+                return []
+            points = store.data[:n].clone()
+            records = [
+                self.engine._latest[self.engine._row_identity[r]] for r in range(n)
+            ]
         k = min(self.n_clusters, n)
         km = StreamingKMeans(k, store.dim, device=str(store.device), seed=17)
-        points = store.data[:n]
         assign = km.fit(points, iters=iters).tolist()
 
         clusters: Dict[int, List[int]] = defaultdict(list)
@@ -41,7 +48,7 @@ class PatternMiner:
             apps: set = set()
             types: Dict[str, int] = defaultdict(int)
             for r in rows:
-                rec = self.engine._latest[self.engine._row_identity[r]]
+                rec = records[r]
                 ids.append(rec["failure_id"])
                 apps.update(rec.get("affected_apps", []))
                 types[rec["failure_type"]] += 1
