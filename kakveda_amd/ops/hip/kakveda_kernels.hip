@@ -34,6 +34,13 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // kept for the next optimization round.
   static const char* ksel = std::getenv("KAKVEDA_KNN_KERNEL");
   const bool use8p = (ksel && std::string(ksel) == "8p") && N >= 4096;
+  // KAKVEDA_KNN_KERNEL=dfr / =rege select the experimental epilogue
+  // variants (7: deferred extraction — measured slower; 8: register
+  // -cached thresholds with eager extraction). Default: EPI_MODE 0.
+  const int epi = ksel ? (std::string(ksel) == "dfr"    ? 7
+                          : std::string(ksel) == "rege" ? 8
+                                                        : 0)
+                       : 0;
 
   const int tile_m = use8p ? BM8 : BM;
   const int tile_n = use8p ? BN8 : BN;
@@ -81,6 +88,18 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
     (void)slab;
+  } else if (epi == 7) {
+    hipLaunchKernelGGL((cosine_topk_partial_t<7>), grid, dim3(THREADS), 0, stream.stream(),
+                       (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       B, N, D, chunk_tiles, nchunks,
+                       (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
+  } else if (epi == 8) {
+    hipLaunchKernelGGL((cosine_topk_partial_t<8>), grid, dim3(THREADS), 0, stream.stream(),
+                       (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       B, N, D, chunk_tiles, nchunks,
+                       (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
   } else {
     hipLaunchKernelGGL(cosine_topk_partial, grid, dim3(THREADS), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),

@@ -229,7 +229,9 @@ __device__ __noinline__ void topk_extract_group(
 //   partial_score/partial_idx: [B][nchunks][KMAX]
 // ---------------------------------------------------------------------------
 // EPI_MODE: 0 full top-k, 1 GEMM-only ablation, 2 pre-check only,
-// 3 full+stats, 4 argmax (k=1 fast path: per-row max, no lists/extraction)
+// 3 full+stats, 4 argmax (k=1 fast path: per-row max, no lists/extraction),
+// 6 ballot-leader extraction, 7 deferred register-threshold epilogue
+// (pre-check in registers, extraction after the tile barrier)
 template <int EPI_MODE, int NKK = 2>  // NKK: 32-deep K steps per LDS stage (2 -> BK=64)
 __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
@@ -301,6 +303,14 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
   __syncthreads();
   int cur = 0;
 
+  // EPI_MODE 7: per-row pruning threshold cached in a register. Lane l of
+  // each wave owns max(list-min, warm) for row wr*64+l of col-half wc, so
+  // the pre-check needs no LDS reads at all (the volatile list reads at
+  // the 16 unrolled call boundaries otherwise serialize ~70 lgkmcnt(0)
+  // waits per tile per wave — measured 650 vs 884 TF full-vs-precheck).
+  float rmin_reg = NEG_INF;
+  (void)rmin_reg;
+
   for (int j = 0; j < tiles_here; ++j) {
     const int col0 = (tile0 + j) * BN;
 
@@ -310,6 +320,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     float warm = NEG_INF;
     if (EPI_MODE != 1 && rowthr != nullptr && row0 + wr * 64 + lane < B)
       warm = dec_f32(rowthr[row0 + wr * 64 + lane]);
+    if constexpr (EPI_MODE == 7 || EPI_MODE == 8) rmin_reg = fmaxf(rmin_reg, warm);
 
     f32x4 acc[4][4];
 #pragma unroll
@@ -359,12 +370,36 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     // row, a 4-step intra-group shfl computes the row-half max from the
     // accumulators; the (rare) qualifying rows extract their candidates in
     // descending order, one leader lane updating the private LDS list.
+    unsigned qmask = 0;
+    (void)qmask;
     if constexpr (EPI_MODE == 1) {
 #pragma unroll
       for (int m = 0; m < 4; ++m)
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
+    } else if constexpr (EPI_MODE == 7) {
+      // pre-check only here (registers + cross-lane ops, no LDS, no
+      // calls); qualifying (m,reg) groups set a bit and extract after
+      // the tile barrier below, where vmcnt is naturally drained so the
+      // noinline callee's conservative entry wait costs nothing.
+      const int colb = col0 + wc * 64 + cl;
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const float rwall = __shfl(rmin_reg, m * 16 + g * 4 + reg, 64);
+          float w0 = (colb + 0 < N) ? acc[m][0][reg] : NEG_INF;
+          float w1 = (colb + 16 < N) ? acc[m][1][reg] : NEG_INF;
+          float w2 = (colb + 32 < N) ? acc[m][2][reg] : NEG_INF;
+          float w3 = (colb + 48 < N) ? acc[m][3][reg] : NEG_INF;
+          float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1)
+            gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
+          if (gmax > rwall) qmask |= 1u << (m * 4 + reg);
+        }
+      }
     } else {
       const int colb = col0 + wc * 64 + cl;
 #pragma unroll
@@ -373,8 +408,18 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
         for (int reg = 0; reg < 4; ++reg) {
           const int row = wr * 64 + m * 16 + g * 4 + reg;
           const int lbase = (wc * BM + row) * KMAX;
-          const float rwarm = __shfl(warm, m * 16 + g * 4 + reg, 64);
-          const float rmin0 = fmaxf(lsc[lbase], rwarm);
+          // EPI_MODE 8 reads the per-row threshold from the register
+          // cache (one cross-lane shuffle) instead of a volatile LDS
+          // read; the 16 unrolled call-boundary reads otherwise cost
+          // ~70 serialized lgkmcnt(0) waits per tile per wave.
+          float rwarm, rmin0;
+          if constexpr (EPI_MODE == 8) {
+            rwarm = __shfl(rmin_reg, m * 16 + g * 4 + reg, 64);
+            rmin0 = rwarm;
+          } else {
+            rwarm = __shfl(warm, m * 16 + g * 4 + reg, 64);
+            rmin0 = fmaxf(lsc[lbase], rwarm);
+          }
           float w0 = (colb + 0 < N) ? acc[m][0][reg] : NEG_INF;
           float w1 = (colb + 16 < N) ? acc[m][1][reg] : NEG_INF;
           float w2 = (colb + 32 < N) ? acc[m][2][reg] : NEG_INF;
@@ -414,14 +459,53 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
                                        w3, colb, N, lane, g, rowthr,
                                        (row0 + row < B) ? row0 + row + 1 : 0);
+              if constexpr (EPI_MODE == 8) qmask = 1;
             }
           }
         }
       }
+      // refresh the register threshold cache from the settled list minima
+      // (wave-local: one coalesced LDS read per lane, before the barrier)
+      if constexpr (EPI_MODE == 8) {
+        if (__any(qmask != 0))
+          rmin_reg = fmaxf(rmin_reg, lsc[(wc * BM + wr * 64 + lane) * KMAX]);
+      }
     }
     __syncthreads();  // next tile's first K-tile staged; lists settled
     cur ^= 1;
+
+    if constexpr (EPI_MODE == 7) {
+      // deferred extraction: prefetch for tile j+1 has completed at the
+      // barrier above, so the callee's entry s_waitcnt vmcnt(0) is free
+      // and the glds stream it would otherwise drain is untouched.
+      if (__any(qmask != 0)) {
+        const int colb = col0 + wc * 64 + cl;
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) {
+            if (qmask & (1u << (m * 4 + reg))) {
+              const int row = wr * 64 + m * 16 + g * 4 + reg;
+              const int lbase = (wc * BM + row) * KMAX;
+              const float rwall = __shfl(rmin_reg, m * 16 + g * 4 + reg, 64);
+              float w0 = (colb + 0 < N) ? acc[m][0][reg] : NEG_INF;
+              float w1 = (colb + 16 < N) ? acc[m][1][reg] : NEG_INF;
+              float w2 = (colb + 32 < N) ? acc[m][2][reg] : NEG_INF;
+              float w3 = (colb + 48 < N) ? acc[m][3][reg] : NEG_INF;
+              topk_extract_group<true>(lsc, lix, lbase, rwall, w0, w1, w2,
+                                       w3, colb, N, lane, g, rowthr,
+                                       (row0 + row < B) ? row0 + row + 1 : 0);
+            }
+          }
+        }
+        // refresh the register threshold from the settled list minima
+        // (slot 0 is each list's min): one coalesced LDS read per lane.
+        rmin_reg = fmaxf(rmin_reg, lsc[(wc * BM + wr * 64 + lane) * KMAX]);
+      }
+    }
   }
+
+  if constexpr (EPI_MODE == 7) __syncthreads();  // deferred lists settled
 
   // write partials: [B][nchunks][KMAX]; merge the two col-half lists
   if (tid < BM) {

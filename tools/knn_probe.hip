@@ -96,6 +96,12 @@ int main(int argc, char** argv) {
     else if (mode == 8)
       hipLaunchKernelGGL((cosine_topk_partial_t<1, 1>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 10)
+      hipLaunchKernelGGL((cosine_topk_partial_t<7>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 11)
+      hipLaunchKernelGGL((cosine_topk_partial_t<8>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
@@ -116,18 +122,19 @@ int main(int argc, char** argv) {
       fprintf(stderr, "launch error (mode %d): %s\n", mode, hipGetErrorString(le));
   };
 
-  const int warm_modes[5] = {0, 4, 9, 5, 1};
-  for (int mi = 0; mi < 5; ++mi) run_mode(warm_modes[mi]);
+  const int NM = 4;
+  const int warm_modes[NM] = {0, 11, 2, 1};
+  for (int mi = 0; mi < NM; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[10] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p"};
-  const int modes[5] = {0, 4, 9, 5, 1};
-  std::vector<std::vector<float>> ms(10);
+  const char* names[12] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128"};
+  const int modes[NM] = {0, 11, 2, 1};
+  std::vector<std::vector<float>> ms(12);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
   for (int it = 0; it < iters; ++it) {
-    for (int mi = 0; mi < 5; ++mi) {
+    for (int mi = 0; mi < NM; ++mi) {
       const int m = modes[mi];
       HIP_CHECK(hipEventRecord(t0));
       run_mode(m);
@@ -154,7 +161,7 @@ int main(int argc, char** argv) {
   printf("stats: stamped=%llu (%.2f/tile) inserts=%llu (%.1f/row/chunk)\n",
          hstats[0], hstats[0] / tiles, hstats[1],
          hstats[1] / ((double)B * nchunks));
-  for (int mi = 0; mi < 5; ++mi) {
+  for (int mi = 0; mi < NM; ++mi) {
     const int m = modes[mi];
     std::sort(ms[m].begin(), ms[m].end());
     const float med = ms[m][ms[m].size() / 2];
