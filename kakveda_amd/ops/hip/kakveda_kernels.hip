@@ -34,13 +34,15 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // kept for the next optimization round.
   static const char* ksel = std::getenv("KAKVEDA_KNN_KERNEL");
   const bool use8p = (ksel && std::string(ksel) == "8p") && N >= 4096;
-  // KAKVEDA_KNN_KERNEL=dfr / =rege select the experimental epilogue
-  // variants (7: deferred extraction — measured slower; 8: register
-  // -cached thresholds with eager extraction). Default: EPI_MODE 0.
-  const int epi = ksel ? (std::string(ksel) == "dfr"    ? 7
-                          : std::string(ksel) == "rege" ? 8
-                                                        : 0)
-                       : 0;
+  // Epilogue selection: default is EPI_MODE 8 (register-cached per-row
+  // thresholds, eager extraction — measured fastest within-probe).
+  // KAKVEDA_KNN_KERNEL=eager -> EPI_MODE 0 (volatile-LDS thresholds),
+  // =dfr -> EPI_MODE 7 (deferred extraction; measured slower, kept for
+  // reference).
+  const int epi = ksel ? (std::string(ksel) == "dfr"     ? 7
+                          : std::string(ksel) == "eager" ? 0
+                                                         : 8)
+                       : 8;
 
   const int tile_m = use8p ? BM8 : BM;
   const int tile_n = use8p ? BN8 : BN;
@@ -73,6 +75,39 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   if (use8p) {
     slab = torch::empty({(long)nchunks * row_tiles * BM8 * BN8},
                         torch::TensorOptions().dtype(torch::kFloat32).device(queries.device()));
+  }
+  // Threshold pre-pass: one cheap launch over the first PRE_TILES*PREG
+  // column tiles fills each row's lists from a ~4k-column sample and
+  // publishes their minima into rowthr, so the main launch's blocks all
+  // start with near-converged pruning thresholds instead of each paying
+  // the bootstrap insert storm (measured: warm thresholds are worth
+  // ~10% end-to-end; the sample scan is ~0.5% extra work, its partial
+  // writes land in slots the main launch overwrites).
+  // grid.x must stay < nchunks so every pre-pass block's partial slot
+  // [row][chunk_id] is in bounds (the main launch overwrites them all)
+  const int preg = std::min(64, nchunks) & ~7;
+  constexpr int PRE_TILES = 8;
+  const bool prepass = !use8p && k > 1 && preg >= 8 && ntiles >= preg * PRE_TILES;
+  if (prepass) {
+    dim3 pgrid(preg, row_tiles);
+    if (epi == 7)
+      hipLaunchKernelGGL((cosine_topk_partial_t<7>), pgrid, dim3(THREADS), 0, stream.stream(),
+                         (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, PRE_TILES, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
+    else if (epi == 0)
+      hipLaunchKernelGGL((cosine_topk_partial_t<0>), pgrid, dim3(THREADS), 0, stream.stream(),
+                         (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, PRE_TILES, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
+    else
+      hipLaunchKernelGGL((cosine_topk_partial_t<8>), pgrid, dim3(THREADS), 0, stream.stream(),
+                         (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, PRE_TILES, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
   }
   if (k == 1 && !use8p) {
     // assignment fast path: per-row argmax epilogue (no lists/extraction)
