@@ -320,7 +320,8 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     float warm = NEG_INF;
     if (EPI_MODE != 1 && rowthr != nullptr && row0 + wr * 64 + lane < B)
       warm = dec_f32(rowthr[row0 + wr * 64 + lane]);
-    if constexpr (EPI_MODE == 7 || EPI_MODE == 8) rmin_reg = fmaxf(rmin_reg, warm);
+    if constexpr (EPI_MODE == 7 || EPI_MODE == 8 || EPI_MODE == 9)
+      rmin_reg = fmaxf(rmin_reg, warm);
 
     f32x4 acc[4][4];
 #pragma unroll
@@ -413,7 +414,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
           // read; the 16 unrolled call-boundary reads otherwise cost
           // ~70 serialized lgkmcnt(0) waits per tile per wave.
           float rwarm, rmin0;
-          if constexpr (EPI_MODE == 8) {
+          if constexpr (EPI_MODE == 8 || EPI_MODE == 9) {
             rwarm = __shfl(rmin_reg, m * 16 + g * 4 + reg, 64);
             rmin0 = rwarm;
           } else {
@@ -455,6 +456,62 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               topk_extract_group_bl<true>(lsc, lix, lbase, rwarm, w0, w1,
                                           w2, w3, colb, N, lane, g, rowthr,
                                           (row0 + row < B) ? row0 + row + 1 : 0);
+            } else if constexpr (EPI_MODE == 9) {
+              // inline single-insert fast path: the group argmax leader
+              // inserts directly (no call, so no callee-entry
+              // s_waitcnt vmcnt(0) draining the staging stream); only
+              // groups with a SECOND candidate above the new min take
+              // the noinline extractor.
+              float b = w0;
+              int bn = 0;
+              if (w1 > b) { b = w1; bn = 1; }
+              if (w2 > b) { b = w2; bn = 2; }
+              if (w3 > b) { b = w3; bn = 3; }
+              float mv = b;
+              int mlane = lane;
+#pragma unroll
+              for (int off = 1; off < 16; off <<= 1) {
+                const float ov = __shfl_xor(mv, off, 64);
+                const int ol = __shfl_xor(mlane, off, 64);
+                if (ov > mv || (ov == mv && ol < mlane)) { mv = ov; mlane = ol; }
+              }
+              float nmn = mv;  // new list min (valid on the leader lane)
+              if (lane == mlane) {
+                int nmp = 0;
+#pragma unroll
+                for (int q = 1; q < KMAX; ++q) {
+                  const float sq = lsc[lbase + q];
+                  if (sq < nmn) { nmn = sq; nmp = q; }
+                }
+                const int gcol = colb + bn * 16;
+                if (nmp == 0) {
+                  lsc[lbase] = mv;
+                  lix[lbase] = gcol;
+                } else {
+                  const int mi2 = lix[lbase + nmp];
+                  lsc[lbase] = nmn;
+                  lix[lbase] = mi2;
+                  lsc[lbase + nmp] = mv;
+                  lix[lbase + nmp] = gcol;
+                }
+                if (bn == 0) w0 = NEG_INF;
+                else if (bn == 1) w1 = NEG_INF;
+                else if (bn == 2) w2 = NEG_INF;
+                else w3 = NEG_INF;
+                if (rowthr != nullptr && nmn > NEG_INF && row0 + row < B)
+                  atomicMax(&rowthr[row0 + row], enc_f32(nmn));
+              }
+              qmask = 1;
+              // does any remaining candidate still beat the new min?
+              const float nmb = fmaxf(__shfl(nmn, mlane, 64), rwarm);
+              float g2 = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+#pragma unroll
+              for (int off = 1; off < 16; off <<= 1)
+                g2 = fmaxf(g2, __shfl_xor(g2, off, 64));
+              if (g2 > nmb)
+                topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
+                                         w3, colb, N, lane, g, rowthr,
+                                         (row0 + row < B) ? row0 + row + 1 : 0);
             } else if constexpr (EPI_MODE != 2) {
               topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
                                        w3, colb, N, lane, g, rowthr,
@@ -466,7 +523,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
       }
       // refresh the register threshold cache from the settled list minima
       // (wave-local: one coalesced LDS read per lane, before the barrier)
-      if constexpr (EPI_MODE == 8) {
+      if constexpr (EPI_MODE == 8 || EPI_MODE == 9) {
         if (__any(qmask != 0))
           rmin_reg = fmaxf(rmin_reg, lsc[(wc * BM + wr * 64 + lane) * KMAX]);
       }

@@ -82,7 +82,7 @@ int main(int argc, char** argv) {
     // modes 12/13: "warm" variants — keep rowthr from the previous
     // iteration (same data, so thresholds converge to the exact per-row
     // k-th best): measures the ideal-threshold-warming ceiling.
-    if (mode != 12 && mode != 13)
+    if (mode != 12 && mode != 13 && mode != 15)
       hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0, 0,
                          rowthr, B);
     if (mode == 0)
@@ -112,6 +112,12 @@ int main(int argc, char** argv) {
     else if (mode == 13)
       hipLaunchKernelGGL((cosine_topk_partial_t<8>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 14)
+      hipLaunchKernelGGL((cosine_topk_partial_t<9>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 15) // warm variant of 9
+      hipLaunchKernelGGL((cosine_topk_partial_t<9>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
@@ -133,13 +139,13 @@ int main(int argc, char** argv) {
   };
 
   const int NM = 6;
-  const int warm_modes[NM] = {0, 11, 12, 13, 2, 1};
+  const int warm_modes[NM] = {0, 11, 14, 15, 13, 2};
   for (int mi = 0; mi < NM; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[14] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128"};
-  const int modes[NM] = {0, 11, 12, 13, 2, 1};
-  std::vector<std::vector<float>> ms(14);
+  const char* names[16] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128"};
+  const int modes[NM] = {0, 11, 14, 15, 13, 2};
+  std::vector<std::vector<float>> ms(16);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
