@@ -320,7 +320,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     float warm = NEG_INF;
     if (EPI_MODE != 1 && rowthr != nullptr && row0 + wr * 64 + lane < B)
       warm = dec_f32(rowthr[row0 + wr * 64 + lane]);
-    if constexpr (EPI_MODE == 7 || EPI_MODE == 8 || EPI_MODE == 9)
+    if constexpr (EPI_MODE >= 7 && EPI_MODE <= 10)
       rmin_reg = fmaxf(rmin_reg, warm);
 
     f32x4 acc[4][4];
@@ -414,7 +414,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
           // read; the 16 unrolled call-boundary reads otherwise cost
           // ~70 serialized lgkmcnt(0) waits per tile per wave.
           float rwarm, rmin0;
-          if constexpr (EPI_MODE == 8 || EPI_MODE == 9) {
+          if constexpr (EPI_MODE == 8 || EPI_MODE == 9 || EPI_MODE == 10) {
             rwarm = __shfl(rmin_reg, m * 16 + g * 4 + reg, 64);
             rmin0 = rwarm;
           } else {
@@ -456,7 +456,24 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               topk_extract_group_bl<true>(lsc, lix, lbase, rwarm, w0, w1,
                                           w2, w3, colb, N, lane, g, rowthr,
                                           (row0 + row < B) ? row0 + row + 1 : 0);
-            } else if constexpr (EPI_MODE == 9) {
+            } else if constexpr (EPI_MODE == 9 || EPI_MODE == 10) {
+              // EPI_MODE 10: during list bootstrap (no threshold yet)
+              // the inline insert would almost always fall through to
+              // the callee anyway — skip straight to it. (A flag, not a
+              // goto: a goto would block loop unrolling and trip the
+              // runtime-indexed-accumulator scratch path.)
+              bool storm = false;
+              if constexpr (EPI_MODE == 10) {
+                if (rwarm <= NEG_INF) {
+                  storm = true;
+                  topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1,
+                                           w2, w3, colb, N, lane, g, rowthr,
+                                           (row0 + row < B) ? row0 + row + 1
+                                                            : 0);
+                  qmask = 1;
+                }
+              }
+              if (!storm) {
               // inline single-insert fast path: the group argmax leader
               // inserts directly (no call, so no callee-entry
               // s_waitcnt vmcnt(0) draining the staging stream); only
@@ -512,6 +529,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
                 topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
                                          w3, colb, N, lane, g, rowthr,
                                          (row0 + row < B) ? row0 + row + 1 : 0);
+              }
             } else if constexpr (EPI_MODE != 2) {
               topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
                                        w3, colb, N, lane, g, rowthr,
@@ -523,7 +541,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
       }
       // refresh the register threshold cache from the settled list minima
       // (wave-local: one coalesced LDS read per lane, before the barrier)
-      if constexpr (EPI_MODE == 8 || EPI_MODE == 9) {
+      if constexpr (EPI_MODE == 8 || EPI_MODE == 9 || EPI_MODE == 10) {
         if (__any(qmask != 0))
           rmin_reg = fmaxf(rmin_reg, lsc[(wc * BM + wr * 64 + lane) * KMAX]);
       }

@@ -118,6 +118,9 @@ int main(int argc, char** argv) {
     else if (mode == 15) // warm variant of 9
       hipLaunchKernelGGL((cosine_topk_partial_t<9>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 16)
+      hipLaunchKernelGGL((cosine_topk_partial_t<10>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
@@ -139,13 +142,13 @@ int main(int argc, char** argv) {
   };
 
   const int NM = 6;
-  const int warm_modes[NM] = {0, 11, 14, 15, 13, 2};
+  const int warm_modes[NM] = {0, 14, 16, 15, 13, 2};
   for (int mi = 0; mi < NM; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[16] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128"};
-  const int modes[NM] = {0, 11, 14, 15, 13, 2};
-  std::vector<std::vector<float>> ms(16);
+  const char* names[17] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128", "faststorm128"};
+  const int modes[NM] = {0, 14, 16, 15, 13, 2};
+  std::vector<std::vector<float>> ms(17);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
