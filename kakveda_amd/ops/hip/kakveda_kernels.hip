@@ -34,18 +34,19 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // kept for the next optimization round.
   static const char* ksel = std::getenv("KAKVEDA_KNN_KERNEL");
   const bool use8p = (ksel && std::string(ksel) == "8p") && N >= 4096;
-  // Epilogue selection: default is EPI_MODE 9 (register-cached per-row
-  // thresholds + inline single-insert fast path with noinline fallback
-  // — measured fastest within-probe: 660 vs 613 TF for EPI_MODE 0 at
-  // B=4096 x N=2M). KAKVEDA_KNN_KERNEL=eager -> 0 (volatile-LDS
-  // thresholds), =rege -> 8 (register thresholds, call-only extraction),
-  // =dfr -> 7 (deferred extraction; measured slower, kept for
-  // reference).
+  // Epilogue selection: default is EPI_MODE 11 (register-cached per-row
+  // thresholds + ballot pre-check + inline single-insert fast path with
+  // noinline fallback — measured fastest within-probe: 745 vs 642 TF
+  // for EPI_MODE 0 at B=4096 x N=2M). KAKVEDA_KNN_KERNEL=eager -> 0
+  // (volatile-LDS thresholds), =rege -> 8 (register thresholds,
+  // call-only extraction), =fast -> 9 (shfl-reduce pre-check), =dfr ->
+  // 7 (deferred extraction; measured slower, kept for reference).
   const int epi = ksel ? (std::string(ksel) == "dfr"     ? 7
                           : std::string(ksel) == "eager" ? 0
                           : std::string(ksel) == "rege"  ? 8
-                                                         : 9)
-                       : 9;
+                          : std::string(ksel) == "fast"  ? 9
+                                                         : 11)
+                       : 11;
 
   const int tile_m = use8p ? BM8 : BM;
   const int tile_n = use8p ? BN8 : BN;
@@ -102,8 +103,8 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        (penv ? penv[0] == '1' : psmall);
   if (prepass) {
     dim3 pgrid(preg, row_tiles);
-    if (epi == 9)
-      hipLaunchKernelGGL((cosine_topk_partial_t<9>), pgrid, dim3(THREADS), 0, stream.stream(),
+    if (epi == 11)
+      hipLaunchKernelGGL((cosine_topk_partial_t<11>), pgrid, dim3(THREADS), 0, stream.stream(),
                          (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
                          pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                          B, N, D, PRE_TILES, nchunks,
@@ -155,6 +156,12 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
   } else if (epi == 9) {
     hipLaunchKernelGGL((cosine_topk_partial_t<9>), grid, dim3(THREADS), 0, stream.stream(),
+                       (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       B, N, D, chunk_tiles, nchunks,
+                       (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
+  } else if (epi == 11) {
+    hipLaunchKernelGGL((cosine_topk_partial_t<11>), grid, dim3(THREADS), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,

@@ -320,7 +320,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     float warm = NEG_INF;
     if (EPI_MODE != 1 && rowthr != nullptr && row0 + wr * 64 + lane < B)
       warm = dec_f32(rowthr[row0 + wr * 64 + lane]);
-    if constexpr (EPI_MODE >= 7 && EPI_MODE <= 10)
+    if constexpr (EPI_MODE >= 7 && EPI_MODE <= 11)
       rmin_reg = fmaxf(rmin_reg, warm);
 
     f32x4 acc[4][4];
@@ -414,7 +414,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
           // read; the 16 unrolled call-boundary reads otherwise cost
           // ~70 serialized lgkmcnt(0) waits per tile per wave.
           float rwarm, rmin0;
-          if constexpr (EPI_MODE == 8 || EPI_MODE == 9 || EPI_MODE == 10) {
+          if constexpr (EPI_MODE >= 8 && EPI_MODE <= 11) {
             rwarm = __shfl(rmin_reg, m * 16 + g * 4 + reg, 64);
             rmin0 = rwarm;
           } else {
@@ -425,11 +425,22 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
           float w1 = (colb + 16 < N) ? acc[m][1][reg] : NEG_INF;
           float w2 = (colb + 32 < N) ? acc[m][2][reg] : NEG_INF;
           float w3 = (colb + 48 < N) ? acc[m][3][reg] : NEG_INF;
-          float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+          // EPI_MODE 11: one ballot replaces the 4-deep shfl_xor max
+          // reduce in the pre-check (same predicate: does ANY lane of
+          // the group beat the threshold) — halves the cross-lane
+          // dependency chain of the hot no-candidate path.
+          bool qual;
+          if constexpr (EPI_MODE == 11) {
+            const float lmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+            qual = (__ballot(lmax > rmin0) & (0xFFFFull << (g * 16))) != 0;
+          } else {
+            float gmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
 #pragma unroll
-          for (int off = 1; off < 16; off <<= 1)
-            gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
-          if (gmax > rmin0) {
+            for (int off = 1; off < 16; off <<= 1)
+              gmax = fmaxf(gmax, __shfl_xor(gmax, off, 64));
+            qual = gmax > rmin0;
+          }
+          if (qual) {
             if constexpr (EPI_MODE == 3) {
               if (cl == 0 && stats) atomicAdd(&stats[0], 1ull);
             }
@@ -456,7 +467,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               topk_extract_group_bl<true>(lsc, lix, lbase, rwarm, w0, w1,
                                           w2, w3, colb, N, lane, g, rowthr,
                                           (row0 + row < B) ? row0 + row + 1 : 0);
-            } else if constexpr (EPI_MODE == 9 || EPI_MODE == 10) {
+            } else if constexpr (EPI_MODE == 9 || EPI_MODE == 10 || EPI_MODE == 11) {
               // EPI_MODE 10: during list bootstrap (no threshold yet)
               // the inline insert would almost always fall through to
               // the callee anyway — skip straight to it. (A flag, not a
@@ -521,11 +532,18 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               qmask = 1;
               // does any remaining candidate still beat the new min?
               const float nmb = fmaxf(__shfl(nmn, mlane, 64), rwarm);
-              float g2 = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+              const float g2l = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
+              bool more;
+              if constexpr (EPI_MODE == 11) {
+                more = (__ballot(g2l > nmb) & (0xFFFFull << (g * 16))) != 0;
+              } else {
+                float g2 = g2l;
 #pragma unroll
-              for (int off = 1; off < 16; off <<= 1)
-                g2 = fmaxf(g2, __shfl_xor(g2, off, 64));
-              if (g2 > nmb)
+                for (int off = 1; off < 16; off <<= 1)
+                  g2 = fmaxf(g2, __shfl_xor(g2, off, 64));
+                more = g2 > nmb;
+              }
+              if (more)
                 topk_extract_group<true>(lsc, lix, lbase, rwarm, w0, w1, w2,
                                          w3, colb, N, lane, g, rowthr,
                                          (row0 + row < B) ? row0 + row + 1 : 0);
@@ -541,7 +559,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
       }
       // refresh the register threshold cache from the settled list minima
       // (wave-local: one coalesced LDS read per lane, before the barrier)
-      if constexpr (EPI_MODE == 8 || EPI_MODE == 9 || EPI_MODE == 10) {
+      if constexpr (EPI_MODE >= 8 && EPI_MODE <= 11) {
         if (__any(qmask != 0))
           rmin_reg = fmaxf(rmin_reg, lsc[(wc * BM + wr * 64 + lane) * KMAX]);
       }

@@ -82,7 +82,7 @@ int main(int argc, char** argv) {
     // modes 12/13: "warm" variants — keep rowthr from the previous
     // iteration (same data, so thresholds converge to the exact per-row
     // k-th best): measures the ideal-threshold-warming ceiling.
-    if (mode != 12 && mode != 13 && mode != 15)
+    if (mode != 12 && mode != 13 && mode != 15 && mode != 19)
       hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0, 0,
                          rowthr, B);
     if (mode == 0)
@@ -121,6 +121,12 @@ int main(int argc, char** argv) {
     else if (mode == 16)
       hipLaunchKernelGGL((cosine_topk_partial_t<10>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 17)
+      hipLaunchKernelGGL((cosine_topk_partial_t<4>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 18 || mode == 19)
+      hipLaunchKernelGGL((cosine_topk_partial_t<11>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
@@ -142,13 +148,13 @@ int main(int argc, char** argv) {
   };
 
   const int NM = 6;
-  const int warm_modes[NM] = {0, 14, 16, 15, 13, 2};
+  const int warm_modes[NM] = {0, 14, 18, 19, 15, 1};
   for (int mi = 0; mi < NM; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[17] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128", "faststorm128"};
-  const int modes[NM] = {0, 14, 16, 15, 13, 2};
-  std::vector<std::vector<float>> ms(17);
+  const char* names[20] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128", "faststorm128", "argmax128", "fastbl128", "fastblwarm128"};
+  const int modes[NM] = {0, 14, 18, 19, 15, 1};
+  std::vector<std::vector<float>> ms(20);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
