@@ -679,7 +679,11 @@ constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 // epilogue (acc quadrants stream to a per-block global slab at tile end;
 // the NEXT tile's windows drain one 32-col slice each, so the top-k
 // maintenance hides under the MFMA pipeline instead of being exposed
-// after the last window).
+// after the last window), 6 = ballot-skip stash (a register ballot
+// pre-check against the rowthr floors marks which waves have ANY
+// qualifying candidate this tile; unflagged waves' stash+drain phases
+// are skipped uniformly — with warm thresholds almost every phase
+// vanishes and the kernel tracks its 1.0 PF GEMM core).
 template <int EPI_MODE>
 __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
@@ -687,7 +691,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     int B, int N, int D, int chunk_tiles, int nchunks,
     unsigned* rowthr = nullptr, unsigned long long* stats = nullptr,
     float* __restrict__ slab = nullptr) {
-  __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4];
+  __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4 + 32];
   char* const smem0 = smem;
   // buffer b in {0,1}: A half h at b*4*HALF8 + h*HALF8; B half h at +2*HALF8
   auto ahalf = [&](int b, int h) -> char* {
@@ -698,6 +702,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
   };
   float* lsc = (float*)(smem + 8 * HALF8);
   int* lix = (int*)(smem + 8 * HALF8 + BM8 * KMAX * 4);
+  int* wflags = (int*)(smem + 8 * HALF8 + 2 * BM8 * KMAX * 4);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -851,6 +856,15 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
       if (rowthr != nullptr && tid < BM8 && row0 + tid < B)
         warm5 = dec_f32(rowthr[row0 + tid]);
     }
+    // EPI_MODE 6: per-lane threshold floors for this wave-half's two
+    // rows (rows beyond B get +inf so clamped-row garbage never flags)
+    float thr0 = NEG_INF, thr1 = NEG_INF;
+    if constexpr (EPI_MODE == 6) {
+      const int r0g = row0 + wr * 128 + lane;
+      thr0 = (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
+      thr1 = (r0g + 64 < B) ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
+                            : 1e38f;
+    }
 
     f32x4 acc[8][4];
 #pragma unroll
@@ -947,6 +961,28 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
     } else {
+      // ---- EPI_MODE 6 pre-check: one register ballot per (m,reg) row
+      // group against the rowthr floors; waves with no qualifying
+      // candidate skip their stash+drain phase below (uniformly — the
+      // flags live in LDS so every wave takes the same branch).
+      if constexpr (EPI_MODE == 6) {
+        unsigned qual = 0;
+#pragma unroll
+        for (int m = 0; m < 8; ++m) {
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) {
+            const int rl = m * 16 + g * 4 + reg;  // 0..127 within half
+            const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
+            const float lmax =
+                fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                      fmaxf(acc[m][2][reg], acc[m][3][reg]));
+            if (__ballot(lmax > thr) & (0xFFFFull << (g * 16)))
+              qual |= 1u;
+          }
+        }
+        if (lane == 0) wflags[wid] = __any(qual != 0) ? 1 : 0;
+        __syncthreads();
+      }
       // ---- top-k epilogue (stash + lane-parallel register-list drain) --
       // The A images of the LAST window's buffer are dead during the
       // epilogue (A(t+2) staging is only issued at window t+1 phase 0),
@@ -965,6 +1001,10 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
       float* stash = (float*)ahalf(t_last & 1, 0);
 #pragma unroll
       for (int phw = 0; phw < 8; ++phw) {
+        if constexpr (EPI_MODE == 6) {
+          // stashing wave's wid = swr*4 + swc; uniform (flags in LDS)
+          if (!wflags[(phw & 1) * 4 + (phw >> 1)]) continue;
+        }
         const int swc = phw >> 1, swr = phw & 1;  // stashing wave
         if (wc == swc && wr == swr) {
 #pragma unroll
