@@ -34,8 +34,10 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // kept for the next optimization round.
   static const char* ksel = std::getenv("KAKVEDA_KNN_KERNEL");
   const bool use8pbl = (ksel && std::string(ksel) == "8pbl") && N >= 4096;
+  const bool use8pq = (ksel && std::string(ksel) == "8pq") && N >= 4096;
   const bool use8p =
-      ((ksel && std::string(ksel) == "8p") || use8pbl) && N >= 4096;
+      ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq) &&
+      N >= 4096;
   // Epilogue selection: default is EPI_MODE 11 (register-cached per-row
   // thresholds + ballot pre-check + inline single-insert fast path with
   // noinline fallback — measured fastest within-probe: 745 vs 642 TF
@@ -137,6 +139,13 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)nullptr, (unsigned long long*)nullptr);
+  } else if (use8pq) {
+    hipLaunchKernelGGL((cosine_topk_partial8p_t<7>), grid, dim3(THREADS8), 0, stream.stream(),
+                       (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       B, N, D, chunk_tiles, nchunks,
+                       (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
+    (void)slab;
   } else if (use8pbl) {
     hipLaunchKernelGGL((cosine_topk_partial8p_t<6>), grid, dim3(THREADS8), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),

@@ -691,7 +691,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     int B, int N, int D, int chunk_tiles, int nchunks,
     unsigned* rowthr = nullptr, unsigned long long* stats = nullptr,
     float* __restrict__ slab = nullptr) {
-  __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4 + 32];
+  __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4 + 32 + BM8 * 4 + 32];
   char* const smem0 = smem;
   // buffer b in {0,1}: A half h at b*4*HALF8 + h*HALF8; B half h at +2*HALF8
   auto ahalf = [&](int b, int h) -> char* {
@@ -703,6 +703,9 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
   float* lsc = (float*)(smem + 8 * HALF8);
   int* lix = (int*)(smem + 8 * HALF8 + BM8 * KMAX * 4);
   int* wflags = (int*)(smem + 8 * HALF8 + 2 * BM8 * KMAX * 4);
+  int* ccnt = (int*)(smem + 8 * HALF8 + 2 * BM8 * KMAX * 4 + 32);
+  unsigned* covf =
+      (unsigned*)(smem + 8 * HALF8 + 2 * BM8 * KMAX * 4 + 32 + BM8 * 4);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -859,7 +862,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     // EPI_MODE 6: per-lane threshold floors for this wave-half's two
     // rows (rows beyond B get +inf so clamped-row garbage never flags)
     float thr0 = NEG_INF, thr1 = NEG_INF;
-    if constexpr (EPI_MODE == 6) {
+    if constexpr (EPI_MODE == 6 || EPI_MODE == 7) {
       const int r0g = row0 + wr * 128 + lane;
       thr0 = (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
       thr1 = (r0g + 64 < B) ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
@@ -961,12 +964,20 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
     } else {
-      // ---- EPI_MODE 6 pre-check: one register ballot per (m,reg) row
-      // group against the rowthr floors; waves with no qualifying
+      // ---- EPI_MODE 6/7 pre-check: one register ballot per (m,reg)
+      // row group against the rowthr floors; waves with no qualifying
       // candidate skip their stash+drain phase below (uniformly — the
       // flags live in LDS so every wave takes the same branch).
-      if constexpr (EPI_MODE == 6) {
-        unsigned qual = 0;
+      bool boot = true;        // stash phases process every row
+      bool run_phases = true;  // whether the stash loop runs at all
+      unsigned qual = 0;       // per-(m,reg) qualifying bits (my groups)
+      (void)boot;
+      (void)run_phases;
+      if constexpr (EPI_MODE == 6 || EPI_MODE == 7) {
+        if constexpr (EPI_MODE == 7) {
+          for (int i = tid; i < BM8; i += THREADS8) ccnt[i] = 0;
+          if (tid < 8) covf[tid] = 0;
+        }
 #pragma unroll
         for (int m = 0; m < 8; ++m) {
 #pragma unroll
@@ -977,11 +988,109 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
                 fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
                       fmaxf(acc[m][2][reg], acc[m][3][reg]));
             if (__ballot(lmax > thr) & (0xFFFFull << (g * 16)))
-              qual |= 1u;
+              qual |= 1u << (m * 4 + reg);
           }
         }
         if (lane == 0) wflags[wid] = __any(qual != 0) ? 1 : 0;
-        __syncthreads();
+        __syncthreads();  // flags + queue counters visible
+      }
+      if constexpr (EPI_MODE == 7) {
+        // EPI_MODE 7: one-pass LDS candidate queue. Qualifying lanes
+        // append (score, col) to their row's slot array; one thread per
+        // row folds them into the shared list. The 8-phase stash loop
+        // runs only for bootstrap tiles (most waves flagged: thresholds
+        // not yet established) and for rows whose queue overflowed —
+        // steady state pays 3 barriers/tile instead of 16.
+        constexpr int CQ8 = 8;
+        const int t_last7 = j * nkt + nkt - 1;
+        float* csc = (float*)ahalf(t_last7 & 1, 0);    // [256][CQ8]
+        int* ccol = (int*)(csc + BM8 * CQ8);           // [256][CQ8]
+        int nfl = 0;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) nfl += wflags[i];
+        boot = nfl >= 6;
+        run_phases = boot;
+        if (!boot) {
+          if (qual) {
+            const int colb = col0 + wc * 64 + cl;
+#pragma unroll
+            for (int m = 0; m < 8; ++m) {
+#pragma unroll
+              for (int reg = 0; reg < 4; ++reg) {
+                if (qual & (1u << (m * 4 + reg))) {
+                  const int row = wr * 128 + m * 16 + g * 4 + reg;
+                  const float thr =
+                      __shfl(m >= 4 ? thr1 : thr0,
+                             (m * 16 + g * 4 + reg) & 63, 64);
+#pragma unroll
+                  for (int n = 0; n < 4; ++n) {
+                    const float v = acc[m][n][reg];
+                    const int gc = colb + n * 16;
+                    if (v > thr && gc < N) {
+                      const int pos = atomicAdd(&ccnt[row], 1);
+                      if (pos < CQ8) {
+                        csc[row * CQ8 + pos] = v;
+                        ccol[row * CQ8 + pos] = gc;
+                      } else {
+                        atomicOr(&covf[row >> 5], 1u << (row & 31));
+                      }
+                    }
+                  }
+                }
+              }
+            }
+          }
+          __syncthreads();  // queue complete
+          if (tid < BM8) {
+            const int row = tid;
+            const int grow = row0 + row;
+            const bool ovf = (covf[row >> 5] >> (row & 31)) & 1;
+            int nq = ccnt[row];
+            if (nq > CQ8) nq = CQ8;
+            if (!ovf && nq > 0 && grow < B) {
+              // in-place LDS list update (slot 0 holds the min): no
+              // register arrays — acc (128 VGPRs) must stay live for
+              // the overflow stash pass, so the drain's footprint has
+              // to be tiny to avoid scratch spill.
+              volatile float* lrow = lsc + row * KMAX;
+              volatile int* irow = lix + row * KMAX;
+              float rmin = lrow[0];
+              bool dirty = false;
+              float mn_out = NEG_INF;
+              for (int i = 0; i < nq; ++i) {
+                const float v = csc[row * CQ8 + i];
+                const int gc = ccol[row * CQ8 + i];
+                if (v > rmin) {
+                  float mn2 = 1e38f;
+                  int m2 = 0;
+#pragma unroll
+                  for (int q = 1; q < KMAX; ++q) {
+                    const float sq = lrow[q];
+                    if (sq < mn2) { mn2 = sq; m2 = q; }
+                  }
+                  if (v <= mn2) {
+                    lrow[0] = v;
+                    irow[0] = gc;
+                  } else {
+                    lrow[0] = mn2;
+                    irow[0] = irow[m2];
+                    lrow[m2] = v;
+                    irow[m2] = gc;
+                  }
+                  const float nmn = fminf(mn2, v);
+                  rmin = nmn;
+                  mn_out = nmn;
+                  dirty = true;
+                }
+              }
+              if (dirty && rowthr != nullptr && mn_out > NEG_INF)
+                atomicMax(&rowthr[grow], enc_f32(mn_out));
+            }
+          }
+          __syncthreads();  // lists settled; covf stable
+          run_phases = (covf[0] | covf[1] | covf[2] | covf[3] | covf[4] |
+                        covf[5] | covf[6] | covf[7]) != 0;
+        }
       }
       // ---- top-k epilogue (stash + lane-parallel register-list drain) --
       // The A images of the LAST window's buffer are dead during the
@@ -1001,7 +1110,10 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
       float* stash = (float*)ahalf(t_last & 1, 0);
 #pragma unroll
       for (int phw = 0; phw < 8; ++phw) {
-        if constexpr (EPI_MODE == 6) {
+        if constexpr (EPI_MODE == 7) {
+          if (!run_phases) break;  // uniform (LDS-derived)
+        }
+        if constexpr (EPI_MODE == 6 || EPI_MODE == 7) {
           // stashing wave's wid = swr*4 + swc; uniform (flags in LDS)
           if (!wflags[(phw & 1) * 4 + (phw >> 1)]) continue;
         }
@@ -1024,6 +1136,14 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
             const int rl = wid * 16 + lane;         // 0..127 within half
             const int row = swr * 128 + rl;
             const int grow = row0 + row;
+            // mode 7: outside bootstrap, the stash pass only repairs
+            // rows whose candidate queue overflowed (queue-inserted
+            // rows are already exact; re-scanning them would be
+            // harmless but wasteful)
+            bool skip_row = false;
+            if constexpr (EPI_MODE == 7)
+              skip_row = !boot && !((covf[row >> 5] >> (row & 31)) & 1);
+            if (!skip_row) {
             const int colq = col0 + swc * 64;
             float ls[KMAX];
             int li[KMAX];
@@ -1083,6 +1203,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
               }
               if (rowthr != nullptr && mn > NEG_INF && grow < B)
                 atomicMax(&rowthr[grow], enc_f32(mn));
+            }
             }
           }
         }
