@@ -675,6 +675,7 @@ constexpr int THREADS8 = 512;
 constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 
 
+
 // EPI_MODE: 0 = stash+drain epilogue, 1 = GEMM only, 5 = slab-deferred
 // epilogue (acc quadrants stream to a per-block global slab at tile end;
 // the NEXT tile's windows drain one 32-col slice each, so the top-k
