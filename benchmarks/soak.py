@@ -31,14 +31,21 @@ def main():
     from kakveda_amd import ops
     from kakveda_amd.gfkb.engine import EmbeddingStore
 
-    store = EmbeddingStore(args.dim, device=device, capacity=args.entries * 2)
+    # capacity for the run's inserts; fill in chunks (a single fp32 randn
+    # of the whole corpus is a 4x-sized temporary and OOMs at 50M+)
+    cap = args.entries + 16_000_000
+    store = EmbeddingStore(args.dim, device=device, capacity=cap)
     gen = torch.Generator(device=device).manual_seed(77)
-    base = torch.randn(args.entries, args.dim, generator=gen, device=device).to(
-        store.dtype
-    )
-    if device == "cuda":
-        ops.l2normalize_(base)
-    store.append(base)
+    fill = 1 << 21
+    for s0 in range(0, args.entries, fill):
+        e0 = min(s0 + fill, args.entries)
+        base = torch.randn(e0 - s0, args.dim, generator=gen, device=device).to(
+            store.dtype
+        )
+        if device == "cuda":
+            ops.l2normalize_(base)
+        store.append(base)
+        del base
 
     lock = threading.Lock()
     stop = threading.Event()
