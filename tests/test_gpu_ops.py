@@ -234,3 +234,41 @@ def test_k1_argmax_fast_path():
     assert torch.allclose(scores[:, 0], ref_s, atol=2e-2, rtol=1e-2)
     gathered = sims.gather(1, idx)
     assert torch.allclose(gathered[:, 0], scores[:, 0], atol=1e-4)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("variant", ["eager", "fast", "rege", "8pbl"])
+def test_kernel_variants_match_default(variant):
+    """Every KAKVEDA_KNN_KERNEL variant must produce the same top-k
+    scores as the default ballot kernel (env is read once per process,
+    so variants run in a subprocess)."""
+    import os
+    import subprocess
+    import sys
+
+    code = (
+        "import torch\n"
+        "from kakveda_amd import ops\n"
+        "q = torch.randn(256, 768, generator=torch.Generator(device='cuda')"
+        ".manual_seed(3), device='cuda')\n"
+        "q = (q / q.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
+        "c = torch.randn(8192, 768, generator=torch.Generator(device='cuda')"
+        ".manual_seed(4), device='cuda')\n"
+        "c = (c / c.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
+        "s, i = ops.cosine_topk(q, c, 5)\n"
+        "torch.cuda.synchronize()\n"
+        "print('CSUM', float(s.double().sum()))\n"
+    )
+    outs = {}
+    for ksel in (None, variant):
+        env = dict(os.environ)
+        env.pop("KAKVEDA_KNN_KERNEL", None)
+        if ksel:
+            env["KAKVEDA_KNN_KERNEL"] = ksel
+        r = subprocess.run(
+            [sys.executable, "-c", code], env=env, capture_output=True,
+            text=True, timeout=300,
+        )
+        assert r.returncode == 0, r.stderr[-1500:]
+        outs[ksel] = float(r.stdout.split("CSUM")[1].strip())
+    assert abs(outs[None] - outs[variant]) < 1e-2, outs
