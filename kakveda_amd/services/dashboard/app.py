@@ -169,6 +169,18 @@ def create_app(
     async def login_page():
         return render("login.html", {})
 
+    @app.get("/register", response_class=HTMLResponse)
+    async def register_page():
+        return render("register.html", {})
+
+    @app.get("/forgot", response_class=HTMLResponse)
+    async def forgot_page():
+        return render("forgot.html", {})
+
+    @app.get("/reset", response_class=HTMLResponse)
+    async def reset_page(token: str = ""):
+        return render("reset.html", {"token": token})
+
     @app.post("/login")
     async def login(request: Request):
         if not ctx.auth_limiter.allow(request.client.host if request.client else "x"):

@@ -544,6 +544,53 @@ def register_features(app: FastAPI, ctx: DashboardContext, u: Dict[str, str]) ->
             )
         return render("scenarios.html", {"scenarios": canned, "recent": recent})
 
+    @app.get("/projects", response_class=HTMLResponse)
+    async def projects_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            projects = s.query(dbm.Project).all()
+            key_counts: Dict[int, int] = {}
+            for kr in s.query(dbm.ProjectApiKey).all():
+                key_counts[kr.project_id] = key_counts.get(kr.project_id, 0) + 1
+            budgets = {
+                b.project_id: b.monthly_usd_micro
+                for b in s.query(dbm.ProjectBudget).all()
+            }
+            return render(
+                "projects.html",
+                {"projects": projects, "key_counts": key_counts, "budgets": budgets},
+            )
+
+    @app.get("/evals", response_class=HTMLResponse)
+    async def evals_page(request: Request):
+        if not ctx.current_user(request):
+            return RedirectResponse("/login", status_code=303)
+        ev = await list_evals()
+        return render("evals.html", {"evals": ev["evals"]})
+
+    @app.get("/admin/users", response_class=HTMLResponse)
+    async def admin_users_page(request: Request):
+        payload = ctx.current_user(request)
+        if not payload or not require_any_roles(payload, ["admin"]):
+            return RedirectResponse("/login", status_code=303)
+        with ctx.Session() as s:
+            roles = {r.id: r.name for r in s.query(dbm.Role).all()}
+            by_user: Dict[int, list] = {}
+            for ur in s.query(dbm.UserRole).all():
+                by_user.setdefault(ur.user_id, []).append(roles.get(ur.role_id, "?"))
+            users = [
+                {
+                    "email": u.email,
+                    "display_name": u.display_name,
+                    "roles": sorted(by_user.get(u.id, [])),
+                    "is_active": u.is_active,
+                    "created_at": u.created_at,
+                }
+                for u in s.query(dbm.User).order_by(dbm.User.id).all()
+            ]
+        return render("admin_users.html", {"users": users})
+
     @app.get("/admin/audit_page", response_class=HTMLResponse)
     async def audit_page(request: Request):
         payload = ctx.current_user(request)

@@ -219,6 +219,58 @@ POST /health/test.</div>{% endfor %}{% endblock %}""",
 {% for r in recent %}<tr><td>{{ r.ts }}</td><td>{{ r.app_id }}</td>
 <td>{{ r.warn_action }}</td><td>{{ "%.2f"|format(r.warn_confidence) }}</td></tr>
 {% endfor %}</table></div>{% endblock %}""",
+    "register.html": """{% extends "base.html" %}{% block content %}
+<div class="card"><h2>Create account</h2>
+<form method="post" action="/register">
+<input name="email" placeholder="email"><br>
+<input name="password" type="password" placeholder="password (8+ chars)"><br>
+<button type="submit">Register</button></form>
+<p><a href="/login">Back to sign in</a></p></div>{% endblock %}""",
+    "forgot.html": """{% extends "base.html" %}{% block content %}
+<div class="card"><h2>Forgot password</h2>
+<form method="post" action="/forgot">
+<input name="email" placeholder="email"><br>
+<button type="submit">Send reset token</button></form>
+<p><a href="/login">Back to sign in</a></p></div>{% endblock %}""",
+    "reset.html": """{% extends "base.html" %}{% block content %}
+<div class="card"><h2>Reset password</h2>
+<form method="post" action="/reset">
+<input name="token" placeholder="reset token" value="{{ token }}"><br>
+<input name="password" type="password" placeholder="new password"><br>
+<button type="submit">Reset</button></form></div>{% endblock %}""",
+    "projects.html": """{% extends "base.html" %}{% block content %}
+<h1>Projects</h1>
+{% for p in projects %}<div class="card"><h3>{{ p.name }}</h3>
+<p>{{ p.description }}</p>
+<p>API keys: {{ key_counts.get(p.id, 0) }} ·
+monthly budget: {{ "%.2f"|format(budgets.get(p.id, 0) / 1000000) }} USD</p>
+<form method="post" action="/api/projects/{{ p.id }}/keys">
+<input name="name" placeholder="key name" value="default">
+<button type="submit">New API key</button></form></div>
+{% else %}<div class="card">No projects yet.</div>{% endfor %}
+<div class="card"><h3>New project</h3>
+<form method="post" action="/api/projects">
+<input name="name" placeholder="name">
+<input name="description" placeholder="description">
+<button type="submit">Create</button></form></div>{% endblock %}""",
+    "admin_users.html": """{% extends "base.html" %}{% block content %}
+<h1>Users</h1>
+<div class="card"><table>
+<tr><th>email</th><th>display</th><th>roles</th><th>active</th><th>created</th></tr>
+{% for u in users %}<tr><td>{{ u.email }}</td><td>{{ u.display_name }}</td>
+<td>{{ u.roles|join(", ") }}</td><td>{{ u.is_active }}</td>
+<td>{{ u.created_at }}</td></tr>{% endfor %}
+</table></div>{% endblock %}""",
+    "evals.html": """{% extends "base.html" %}{% block content %}
+<h1>Evaluations</h1>
+<div class="card"><table>
+<tr><th>ts</th><th>name</th><th>dataset</th><th>pass rate</th><th>p50 ms</th></tr>
+{% for e in evals %}<tr><td>{{ e.ts }}</td>
+<td><a href="/evals/{{ e.id }}">{{ e.name }}</a></td>
+<td>{{ e.dataset_id }}</td>
+<td>{{ e.summary.get("pass_rate", "-") }}</td>
+<td>{{ e.summary.get("p50_ms", "-") }}</td></tr>{% endfor %}
+</table></div>{% endblock %}""",
 }
 
 _env = Environment(loader=DictLoader(_TEMPLATES), autoescape=select_autoescape(["html"]))

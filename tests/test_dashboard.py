@@ -278,12 +278,15 @@ async def test_html_pages(tmp_path):
     async with _client(cluster) as client:
         login_page = await client.get("/login")
         assert login_page.status_code == 200 and "Sign in" in login_page.text
+        for public in ("/register", "/forgot", "/reset"):
+            resp = await client.get(public)
+            assert resp.status_code == 200, public
         await _login(client)
         home = await client.get("/")
         assert "Failure Intelligence" in home.text
         for page in ("/warnings", "/runs", "/playground", "/agents",
                      "/datasets", "/prompts", "/experiments", "/health",
-                     "/scenarios"):
+                     "/scenarios", "/projects", "/evals", "/admin/users"):
             resp = await client.get(page)
             assert resp.status_code == 200, page
     await cluster.aclose()
