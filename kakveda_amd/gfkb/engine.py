@@ -214,7 +214,10 @@ class GfkbEngine:
             scores, idx = self.store.search(q, min(k, self.store.count))
         out: List[FailureMatch] = []
         for s, i in zip(scores[0].tolist(), idx[0].tolist()):
-            if i < 0:
+            if i < 0 or i >= len(self._row_identity):
+                # rows the engine has no identity for (e.g. a pre-loaded
+                # corpus adopted under the engine) can win the similarity
+                # search but cannot be reported as failures
                 continue
             rec = self._latest[self._row_identity[i]]
             if failure_type and rec["failure_type"] != failure_type:
