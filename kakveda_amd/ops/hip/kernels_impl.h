@@ -228,10 +228,17 @@ __device__ __noinline__ void topk_extract_group(
 //   grid.x = nchunks, grid.y = ceil(B/128), block = 256 threads.
 //   partial_score/partial_idx: [B][nchunks][KMAX]
 // ---------------------------------------------------------------------------
-// EPI_MODE: 0 full top-k, 1 GEMM-only ablation, 2 pre-check only,
-// 3 full+stats, 4 argmax (k=1 fast path: per-row max, no lists/extraction),
-// 6 ballot-leader extraction, 7 deferred register-threshold epilogue
-// (pre-check in registers, extraction after the tile barrier)
+// EPI_MODE: 0 full top-k (volatile-LDS thresholds, immediate extraction),
+// 1 GEMM-only ablation, 2 pre-check only (NOTE: its empty taken-branch
+// lets the compiler dead-code-eliminate the MFMAs — not a valid ceiling;
+// use 1), 3 full+stats, 4 argmax (k=1 fast path: per-row max, no
+// lists/extraction), 6 ballot-leader extraction, 7 deferred
+// register-threshold epilogue (extraction after the tile barrier;
+// measured slower), 8 register-cached thresholds + call-only extraction,
+// 9 = 8 + inline single-insert fast path, 10 = 9 + bootstrap bypass
+// (parity), 11 = 9 with a masked-__ballot pre-check instead of the
+// 4-deep shfl_xor max reduce — the PRODUCTION DEFAULT (754 TF vs 642
+// for mode 0 at B=4096 x N=2M; see profiles/knn_kernel_history.md).
 template <int EPI_MODE, int NKK = 2>  // NKK: 32-deep K steps per LDS stage (2 -> BK=64)
 __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
@@ -683,8 +690,12 @@ constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 // after the last window), 6 = ballot-skip stash (a register ballot
 // pre-check against the rowthr floors marks which waves have ANY
 // qualifying candidate this tile; unflagged waves' stash+drain phases
-// are skipped uniformly — with warm thresholds almost every phase
-// vanishes and the kernel tracks its 1.0 PF GEMM core).
+// are skipped uniformly — best 8p variant: 686 TF vs 420 for mode 0),
+// 7 = per-row LDS candidate queue + single drain with the stash loop
+// kept for bootstrap/overflow repair (correct but spills ~490 B/lane:
+// the accumulators must stay live across the repair path — see
+// profiles/knn_kernel_history.md for why a queue-only version was
+// abandoned at the 256-VGPR/2-wave cap).
 template <int EPI_MODE>
 __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
