@@ -327,7 +327,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
     float warm = NEG_INF;
     if (EPI_MODE != 1 && rowthr != nullptr && row0 + wr * 64 + lane < B)
       warm = dec_f32(rowthr[row0 + wr * 64 + lane]);
-    if constexpr (EPI_MODE >= 7 && EPI_MODE <= 11)
+    if constexpr (EPI_MODE >= 7 && EPI_MODE <= 12)
       rmin_reg = fmaxf(rmin_reg, warm);
 
     f32x4 acc[4][4];
@@ -358,12 +358,17 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           bfrag[n] = read_frag_n<NKK>(bbuf(cur), wc * 64 + n * 16 + cl, slot);
+        // EPI_MODE 12 experiment: raise wave priority through the MFMA
+        // burst (8p-style s_setprio) so the co-resident wave's VALU work
+        // interleaves under it
+        if constexpr (EPI_MODE == 12) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n)
             acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag[m], bfrag[n], acc[m][n], 0, 0, 0);
+        if constexpr (EPI_MODE == 12) __builtin_amdgcn_s_setprio(0);
       }
       if (kt + 1 < nkt) {
         __syncthreads();  // drains prefetch glds; guards buffer reuse
@@ -437,7 +442,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
           // the group beat the threshold) — halves the cross-lane
           // dependency chain of the hot no-candidate path.
           bool qual;
-          if constexpr (EPI_MODE == 11) {
+          if constexpr (EPI_MODE == 11 || EPI_MODE == 12) {
             const float lmax = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
             qual = (__ballot(lmax > rmin0) & (0xFFFFull << (g * 16))) != 0;
           } else {
@@ -474,7 +479,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               topk_extract_group_bl<true>(lsc, lix, lbase, rwarm, w0, w1,
                                           w2, w3, colb, N, lane, g, rowthr,
                                           (row0 + row < B) ? row0 + row + 1 : 0);
-            } else if constexpr (EPI_MODE == 9 || EPI_MODE == 10 || EPI_MODE == 11) {
+            } else if constexpr (EPI_MODE >= 9 && EPI_MODE <= 12) {
               // EPI_MODE 10: during list bootstrap (no threshold yet)
               // the inline insert would almost always fall through to
               // the callee anyway — skip straight to it. (A flag, not a
@@ -541,7 +546,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
               const float nmb = fmaxf(__shfl(nmn, mlane, 64), rwarm);
               const float g2l = fmaxf(fmaxf(w0, w1), fmaxf(w2, w3));
               bool more;
-              if constexpr (EPI_MODE == 11) {
+              if constexpr (EPI_MODE == 11 || EPI_MODE == 12) {
                 more = (__ballot(g2l > nmb) & (0xFFFFull << (g * 16))) != 0;
               } else {
                 float g2 = g2l;
@@ -566,7 +571,7 @@ __global__ __launch_bounds__(THREADS, 4 - NKK) void cosine_topk_partial_t(
       }
       // refresh the register threshold cache from the settled list minima
       // (wave-local: one coalesced LDS read per lane, before the barrier)
-      if constexpr (EPI_MODE >= 8 && EPI_MODE <= 11) {
+      if constexpr (EPI_MODE >= 8 && EPI_MODE <= 12) {
         if (__any(qmask != 0))
           rmin_reg = fmaxf(rmin_reg, lsc[(wc * BM + wr * 64 + lane) * KMAX]);
       }

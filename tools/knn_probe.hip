@@ -133,6 +133,9 @@ int main(int argc, char** argv) {
     else if (mode == 21)
       hipLaunchKernelGGL((cosine_topk_partial8p_t<7>), grid2, dim3(THREADS8), 0, 0,
                          Q, C, pscore2, pidx2, B, (int)N, D, chunk_tiles2, nchunks2, rowthr, (unsigned long long*)nullptr);
+    else if (mode == 22)
+      hipLaunchKernelGGL((cosine_topk_partial_t<12>), grid, dim3(THREADS), 0, 0,
+                         Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
     else if (mode == 6)
       hipLaunchKernelGGL((cosine_topk_partial_t<6>), grid, dim3(THREADS), 0, 0,
                          Q, C, pscore, pidx, B, (int)N, D, chunk_tiles, nchunks, rowthr, (unsigned long long*)nullptr);
@@ -154,13 +157,13 @@ int main(int argc, char** argv) {
   };
 
   const int NM = 6;
-  const int warm_modes[NM] = {3, 20, 21, 5, 18, 1};
+  const int warm_modes[NM] = {18, 22, 1, 0, 14, 20};
   for (int mi = 0; mi < NM; ++mi) run_mode(warm_modes[mi]);
   HIP_CHECK(hipDeviceSynchronize());
 
-  const char* names[22] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128", "faststorm128", "argmax128", "fastbl128", "fastblwarm128", "fastbl8p", "queue8p"};
-  const int modes[NM] = {3, 20, 21, 5, 18, 1};
-  std::vector<std::vector<float>> ms(22);
+  const char* names[23] = {"full128", "gemm128", "precheck128", "-", "slab8p", "gemm8p", "full128-bl", "full128-bk32", "gemm128-bk32", "precheck8p", "dfr128", "rege128", "fullwarm128", "regewarm128", "fast128", "fastwarm128", "faststorm128", "argmax128", "fastbl128", "fastblwarm128", "fastbl8p", "queue8p", "fastblprio128"};
+  const int modes[NM] = {18, 22, 1, 0, 14, 20};
+  std::vector<std::vector<float>> ms(23);
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
