@@ -28,10 +28,11 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
 
   // kernel selection. Default: the 128x128 2-blocks/CU kernel (fastest
   // END-TO-END: its epilogue hides under the sibling block's MFMAs).
-  // KAKVEDA_KNN_KERNEL=8p opts into the experimental 256x256 8-phase
-  // pipeline whose GEMM core measures ~1.0 PF but whose epilogue is
-  // exposed at 1 block/CU (and which still fails two numerics tests) —
-  // kept for the next optimization round.
+  // KAKVEDA_KNN_KERNEL=8p/8pbl/8pq opt into the experimental 256x256
+  // 8-phase pipeline (stash / ballot-skip / queue epilogue; all exact,
+  // GPU-suite-covered) whose GEMM core measures ~1.0 PF but whose
+  // epilogue is still partially exposed at 1 block/CU — the remaining
+  // round-2 item (ROUND2.md).
   static const char* ksel = std::getenv("KAKVEDA_KNN_KERNEL");
   const bool use8pbl = (ksel && std::string(ksel) == "8pbl") && N >= 4096;
   const bool use8pq = (ksel && std::string(ksel) == "8pq") && N >= 4096;
