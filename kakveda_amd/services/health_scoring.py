@@ -56,10 +56,20 @@ def create_app(
 
     @app.on_event("startup")
     async def _startup():
-        try:
-            await subscribe()
-        except Exception:
-            pass
+        # the bus may come up after us (all services start concurrently
+        # under `kakveda up`): retry in the background until subscribed;
+        # /subscribe dedups, so LocalCluster's direct call stays safe
+        import asyncio as _aio
+
+        async def _retry():
+            for _ in range(30):
+                try:
+                    await subscribe()
+                    return
+                except Exception:
+                    await _aio.sleep(1.0)
+
+        _aio.get_event_loop().create_task(_retry())
 
     @app.post("/events/failure")
     async def on_failure(event: dict):
