@@ -326,10 +326,19 @@ def create_app(
 
     @app.on_event("startup")
     async def _startup():
-        try:
-            await subscribe()
-        except Exception:
-            pass  # bus may not be up yet (reference app.py:1350-1352)
+        # bus may come up after us under `kakveda up` (all services start
+        # concurrently): retry in the background; /subscribe dedups
+        import asyncio as _aio
+
+        async def _retry():
+            for _ in range(30):
+                try:
+                    await subscribe()
+                    return
+                except Exception:
+                    await _aio.sleep(1.0)
+
+        _aio.get_event_loop().create_task(_retry())
 
     @app.post("/events/trace")
     async def on_trace(event: dict):
