@@ -63,6 +63,9 @@ def _require_ext():
 # cosine top-k
 # ---------------------------------------------------------------------------
 
+#: per-row candidate capacity of the fused HIP kernel (kakveda_kernels.hip)
+KMAX = 8
+
 def cosine_topk(
     queries: torch.Tensor, corpus: torch.Tensor, k: int, valid_n: Optional[int] = None
 ) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -72,17 +75,27 @@ def cosine_topk(
     inner product is the cosine. Returns (scores f32 [B, k], idx i64 [B, k])
     sorted descending per row. ``valid_n`` limits the search to the first
     valid_n corpus rows (the live prefix of a preallocated store).
+
+    The HIP kernel keeps at most KMAX=8 candidates per row; a request with
+    k > 8 on GPU returns the true top-8 padded with (-inf, -1) so CPU and
+    GPU agree on the leading 8 columns instead of the GPU path raising.
     """
     n = int(valid_n) if valid_n is not None else corpus.shape[0]
+    B = queries.shape[0]
     if n <= 0:
-        B = queries.shape[0]
         return (
             torch.full((B, k), float("-inf"), dtype=torch.float32, device=queries.device),
             torch.full((B, k), -1, dtype=torch.int64, device=queries.device),
         )
     if queries.device.type == "cuda":
         ext = _require_ext()
-        scores, idx = ext.cosine_topk(queries, corpus, int(k), n)
+        kk = min(int(k), KMAX)
+        scores, idx = ext.cosine_topk(queries, corpus, kk, n)
+        if kk < k:
+            pad_s = torch.full((B, k - kk), float("-inf"), dtype=scores.dtype, device=scores.device)
+            pad_i = torch.full((B, k - kk), -1, dtype=idx.dtype, device=idx.device)
+            scores = torch.cat([scores, pad_s], dim=1)
+            idx = torch.cat([idx, pad_i], dim=1)
         return scores, idx
     return cosine_topk_ref(queries, corpus, k, n)
 

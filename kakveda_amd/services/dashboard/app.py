@@ -9,6 +9,7 @@ from __future__ import annotations
 
 import datetime as dt
 import json
+import os
 import time
 import uuid
 from typing import Any, Dict, List, Optional
@@ -77,8 +78,13 @@ def bootstrap(ctx: DashboardContext) -> None:
                 )
                 s.add(user)
                 s.flush()
-            elif not verify_password(password, user.password_hash):
-                user.password_hash = hash_password(password)  # self-repair
+            elif not verify_password(password, user.password_hash) and os.environ.get(
+                "DASHBOARD_BOOTSTRAP_FORCE_PASSWORDS", "0"
+            ) == "1":
+                # password self-repair is opt-in (matching the reference's
+                # DASHBOARD_BOOTSTRAP_FORCE_PASSWORDS gate) so a rotated
+                # admin password is never silently reverted on restart
+                user.password_hash = hash_password(password)
             have = {
                 ur.role_id
                 for ur in s.query(dbm.UserRole).filter_by(user_id=user.id).all()
@@ -133,11 +139,31 @@ def create_app(
 
     # -- middleware: request id, duration log, security headers, CSRF cookie --
 
+    # JSON API routes require a logged-in user (reference guards these with
+    # Depends(require_login)); the exceptions carry their own guard:
+    # agent self-registration/heartbeat and external ingest are project
+    # API-key-authenticated, /api/me returns its own 401 contract, and
+    # /events/* are internal event-bus callbacks (see SECURITY.md).
+    PUBLIC_API_PATHS = {
+        "/api/me",
+        "/api/agents/register",
+        "/api/agents/heartbeat",
+        "/api/ingest/run",
+    }
+
     @app.middleware("http")
     async def _middleware(request: Request, call_next):
         rid = request.headers.get("X-Request-Id") or uuid.uuid4().hex
         t0 = time.perf_counter()
-        response: Response = await call_next(request)
+        path = request.url.path
+        if (
+            path.startswith("/api/") or path == "/eval/run"
+        ) and path not in PUBLIC_API_PATHS and not ctx.current_user(request):
+            response: Response = JSONResponse(
+                {"ok": False, "error": "auth_required"}, status_code=401
+            )
+        else:
+            response = await call_next(request)
         dur = (time.perf_counter() - t0) * 1000.0
         response.headers["X-Request-Id"] = rid
         response.headers["X-Content-Type-Options"] = "nosniff"

@@ -55,12 +55,17 @@ def create_app(
         )
     app.state.engine = engine
 
+    # handlers are plain `def` so FastAPI runs them in its threadpool: the
+    # engine is synchronous (GPU kernels, file appends) and must not block
+    # the event loop — a slow match would stall every concurrent request
+    # including /healthz (matches the reference's sync-handler behaviour)
+
     @app.get("/failures")
-    async def list_failures():
+    def list_failures():
         return {"failures": engine.list_failures()}
 
     @app.post("/failures/match", response_model=FailureMatchResponse)
-    async def match(req: FailureMatchRequest):
+    def match(req: FailureMatchRequest):
         from kakveda_amd.core.metrics import observe_gfkb
 
         matches = engine.match(req.signature_text, failure_type=req.failure_type)
@@ -68,7 +73,7 @@ def create_app(
         return FailureMatchResponse(matches=matches)
 
     @app.post("/failures/upsert")
-    async def upsert(req: UpsertFailureRequest):
+    def upsert(req: UpsertFailureRequest):
         rec, created = engine.upsert_failure(
             failure_type=req.failure_type,
             signature_text=req.signature_text,
@@ -81,11 +86,11 @@ def create_app(
         return {"ok": True, "created": created, "failure": rec}
 
     @app.get("/patterns")
-    async def list_patterns():
+    def list_patterns():
         return {"patterns": engine.list_patterns()}
 
     @app.post("/patterns/upsert")
-    async def upsert_pattern(req: UpsertPatternRequest):
+    def upsert_pattern(req: UpsertPatternRequest):
         rec, created = engine.upsert_pattern(
             name=req.name,
             failure_ids=req.failure_ids,
@@ -95,7 +100,7 @@ def create_app(
         return {"ok": True, "created": created, "pattern": rec}
 
     @app.get("/healthz")
-    async def healthz():
+    def healthz():
         from kakveda_amd.core.metrics import observe_gfkb
 
         observe_gfkb(engine.store.count, len(engine.failures))

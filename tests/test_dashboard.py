@@ -307,3 +307,66 @@ async def test_failure_detail_version_addressing(tmp_path):
         missing = await client.get("/failure/F-9999")
         assert missing.status_code == 404
     await cluster.aclose()
+
+
+async def test_api_routes_require_auth(tmp_path):
+    """JSON API routes reject anonymous callers with 401 (ADVICE round 1);
+    API-key-guarded and contract-carrying endpoints stay reachable."""
+    cluster = await _cluster(tmp_path)
+    async with _client(cluster) as client:
+        protected = [
+            ("get", "/api/runs", None),
+            ("get", "/api/datasets", None),
+            ("post", "/api/datasets", {"name": "x"}),
+            ("post", "/api/playground/run", {"prompt": "p"}),
+            ("post", "/eval/run", {"dataset_id": 1}),
+            ("post", "/api/prompts", {"name": "p", "content": "c"}),
+            ("post", "/api/experiments", {"name": "e"}),
+            ("get", "/api/agents", None),
+            ("get", "/api/projects", None),
+            ("get", "/api/warnings", None),
+        ]
+        for method, path, body in protected:
+            resp = await getattr(client, method)(
+                path, **({"json": body} if body is not None else {})
+            )
+            assert resp.status_code == 401, (path, resp.status_code)
+            assert resp.json()["error"] == "auth_required", path
+        # API-key-guarded endpoints answer with their own contract, not 401-auth
+        resp = await client.post("/api/ingest/run", json={"prompt": "p"})
+        assert resp.json()["error"] != "auth_required"
+        resp = await client.post("/api/agents/heartbeat", json={})
+        assert resp.json()["error"] == "invalid_api_key"
+        # and a logged-in caller passes the guard
+        await _login(client)
+        resp = await client.get("/api/runs")
+        assert resp.status_code == 200
+    await cluster.aclose()
+
+
+async def test_bootstrap_password_repair_gated(tmp_path, monkeypatch):
+    """Rotating a demo password survives a restart unless the operator
+    opts into DASHBOARD_BOOTSTRAP_FORCE_PASSWORDS=1."""
+    from kakveda_amd.services.dashboard import db as dbm
+    from kakveda_amd.services.dashboard.app import bootstrap
+    from kakveda_amd.services.dashboard.auth import hash_password, verify_password
+
+    cluster = await _cluster(tmp_path)
+    ctx = cluster.dashboard.state.ctx
+    with ctx.Session() as s:
+        admin = s.query(dbm.User).filter_by(email="admin@kakveda.local").first()
+        admin.password_hash = hash_password("rotated-password-1")
+        s.commit()
+
+    monkeypatch.delenv("DASHBOARD_BOOTSTRAP_FORCE_PASSWORDS", raising=False)
+    bootstrap(ctx)  # simulated restart: must NOT revert
+    with ctx.Session() as s:
+        admin = s.query(dbm.User).filter_by(email="admin@kakveda.local").first()
+        assert verify_password("rotated-password-1", admin.password_hash)
+
+    monkeypatch.setenv("DASHBOARD_BOOTSTRAP_FORCE_PASSWORDS", "1")
+    bootstrap(ctx)  # explicit opt-in: reverts to the documented default
+    with ctx.Session() as s:
+        admin = s.query(dbm.User).filter_by(email="admin@kakveda.local").first()
+        assert verify_password("admin123", admin.password_hash)
+    await cluster.aclose()
