@@ -105,7 +105,7 @@ def main():
     best = None
     for s0 in range(0, n_final, step):
         e0 = min(s0 + step, n_final)
-        sims = q.float() @ store.data[s0:e0].float().t()
+        sims = q.float() @ store.row_range(s0, e0).float().t()
         ts, _ = torch.topk(sims, min(5, e0 - s0), dim=1)
         best = ts if best is None else torch.cat([best, ts], dim=1)
         if best.shape[1] > 64:

@@ -30,7 +30,7 @@ class PatternMiner:
             n = store.count
             if n < 2:
                 return []
-            points = store.data[:n].clone()
+            points = store.row_range(0, n).clone()
             records = [
                 self.engine._latest[self.engine._row_identity[r]] for r in range(n)
             ]
