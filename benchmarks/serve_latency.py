@@ -78,27 +78,47 @@ async def run(args) -> dict:
         # warmup
         for _ in range(5):
             await tx.post("http://warning-policy:8104/warn", json=body)
+
         lat = []
-        t0 = time.perf_counter()
         matched = 0
-        while time.perf_counter() - t0 < args.seconds:
-            ts = time.perf_counter()
-            r = await tx.post("http://warning-policy:8104/warn", json=body)
-            lat.append(time.perf_counter() - ts)
-            if r.json().get("references"):
-                matched += 1
+        stop_at = time.perf_counter() + args.seconds
+
+        async def client_loop():
+            nonlocal matched
+            while time.perf_counter() < stop_at:
+                ts = time.perf_counter()
+                r = await tx.post("http://warning-policy:8104/warn", json=body)
+                lat.append(time.perf_counter() - ts)
+                if r.json().get("references"):
+                    matched += 1
+
+        t0 = time.perf_counter()
+        await asyncio.gather(*(client_loop() for _ in range(args.clients)))
+        wall = time.perf_counter() - t0
         lat_ms = sorted(x * 1000 for x in lat)
+        gfkb = tx.local_app("http://gfkb:8101")
+        batcher = getattr(gfkb.state, "batcher", None)
         return {
             "metric": "serve_warn_latency",
             "unit": "ms",
+            "clients": args.clients,
             "requests": len(lat),
-            "rps": round(len(lat) / (time.perf_counter() - t0), 1),
+            "rps": round(len(lat) / wall, 1),
             "p50_ms": round(lat_ms[len(lat_ms) // 2], 2),
             "p99_ms": round(lat_ms[max(0, int(len(lat_ms) * 0.99) - 1)], 2),
             "matched": matched,
             "entries": n,
             "device": device,
             "data": "synthetic",
+            "batcher": (
+                {
+                    "batches": batcher.batches,
+                    "requests": batcher.requests,
+                    "avg_batch": round(batcher.requests / max(1, batcher.batches), 2),
+                }
+                if batcher is not None
+                else None
+            ),
         }
 
 
@@ -107,6 +127,8 @@ def main():
     ap.add_argument("--entries", type=int, default=10_000_000)
     ap.add_argument("--seconds", type=float, default=15.0)
     ap.add_argument("--dim", type=int, default=768)
+    ap.add_argument("--clients", type=int, default=1,
+                    help="concurrent client loops (micro-batcher coalesces them)")
     args = ap.parse_args()
     out = asyncio.run(run(args))
     print(json.dumps(out))

@@ -10,6 +10,8 @@ instead of a per-request TF-IDF refit.
 from __future__ import annotations
 
 import os
+import queue
+import threading
 from typing import Any, Dict, List, Optional
 
 from fastapi import FastAPI
@@ -21,6 +23,75 @@ from kakveda_amd.core.schemas import (
     Severity,
 )
 from kakveda_amd.gfkb.engine import GfkbEngine
+
+
+class MatchBatcher:
+    """Coalesces concurrent /failures/match requests into one fused-kernel
+    launch (serving concurrency).
+
+    Adaptive batching with zero added idle latency: the collector thread
+    blocks for the first request, then drains whatever else is queued at
+    that instant (requests naturally accumulate while the previous
+    batch's kernel is in flight) — a single client sees one thread-hop,
+    N concurrent clients share one engine.match_batch launch instead of
+    serialising N kernel launches behind the engine lock.
+    """
+
+    def __init__(self, engine: GfkbEngine, max_batch: int = 256):
+        self.engine = engine
+        self.max_batch = max_batch
+        self._q: "queue.SimpleQueue" = queue.SimpleQueue()
+        self._thread: Optional[threading.Thread] = None
+        self._start_lock = threading.Lock()
+        self.batches = 0  # observability: launches issued
+        self.requests = 0  # requests served through the batcher
+
+    def _ensure_thread(self) -> None:
+        if self._thread is None or not self._thread.is_alive():
+            with self._start_lock:
+                if self._thread is None or not self._thread.is_alive():
+                    self._thread = threading.Thread(
+                        target=self._loop, name="gfkb-match-batcher", daemon=True
+                    )
+                    self._thread.start()
+
+    def match(self, signature_text: str, failure_type: Optional[str] = None):
+        self._ensure_thread()
+        item: Dict[str, Any] = {
+            "text": signature_text,
+            "ftype": failure_type,
+            "ev": threading.Event(),
+            "res": None,
+            "err": None,
+        }
+        self._q.put(item)
+        if not item["ev"].wait(timeout=120.0):
+            raise TimeoutError("match batcher timed out")
+        if item["err"] is not None:
+            raise item["err"]
+        return item["res"]
+
+    def _loop(self) -> None:
+        while True:
+            batch = [self._q.get()]
+            while len(batch) < self.max_batch:
+                try:
+                    batch.append(self._q.get_nowait())
+                except queue.Empty:
+                    break
+            try:
+                results = self.engine.match_batch(
+                    [b["text"] for b in batch], [b["ftype"] for b in batch]
+                )
+                for b, r in zip(batch, results):
+                    b["res"] = r
+            except Exception as exc:  # surfaced to every waiting request
+                for b in batch:
+                    b["err"] = exc
+            self.batches += 1
+            self.requests += len(batch)
+            for b in batch:
+                b["ev"].set()
 
 
 class UpsertFailureRequest(BaseModel):
@@ -54,6 +125,13 @@ def create_app(
             data_dir=data_dir or os.environ.get("DATA_DIR", "/app/data"), device=dev
         )
     app.state.engine = engine
+    # micro-batcher on by default; KAKVEDA_MATCH_BATCHER=0 opts out
+    batcher = (
+        MatchBatcher(engine)
+        if os.environ.get("KAKVEDA_MATCH_BATCHER", "1") == "1"
+        else None
+    )
+    app.state.batcher = batcher
 
     # handlers are plain `def` so FastAPI runs them in its threadpool: the
     # engine is synchronous (GPU kernels, file appends) and must not block
@@ -68,7 +146,10 @@ def create_app(
     def match(req: FailureMatchRequest):
         from kakveda_amd.core.metrics import observe_gfkb
 
-        matches = engine.match(req.signature_text, failure_type=req.failure_type)
+        if batcher is not None:
+            matches = batcher.match(req.signature_text, failure_type=req.failure_type)
+        else:
+            matches = engine.match(req.signature_text, failure_type=req.failure_type)
         observe_gfkb(engine.store.count, len(engine.failures))
         return FailureMatchResponse(matches=matches)
 

@@ -26,6 +26,10 @@ class Transport:
         self._local[base_url.rstrip("/")] = app
         self._clients.pop(base_url.rstrip("/"), None)
 
+    def local_app(self, base_url: str) -> Optional[Any]:
+        """The locally-registered ASGI app for a base URL, if any."""
+        return self._local.get(base_url.rstrip("/"))
+
     def _client(self, url: str) -> tuple[httpx.AsyncClient, str]:
         for base, app in self._local.items():
             if url.startswith(base):

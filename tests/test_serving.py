@@ -1,0 +1,148 @@
+"""Concurrent serving path: batched matching + the service micro-batcher
+(VERDICT round 1 weak #2)."""
+
+import threading
+import time
+
+from kakveda_amd.gfkb.engine import GfkbEngine
+from kakveda_amd.services.gfkb_service import MatchBatcher
+
+
+def _engine(tmp_path):
+    eng = GfkbEngine(data_dir=str(tmp_path), device="cpu", dim=256, hash_dim=4096)
+    for i in range(8):
+        eng.upsert_failure(
+            "HALLUCINATION_CITATION" if i % 2 == 0 else "TIMEOUT",
+            f"intent_tags:intent:citations_required | prompt_hint:probe {i} | "
+            "tools: | env_keys:e2e",
+            {"i": i},
+            app_id=f"app-{i % 3}",
+        )
+    return eng
+
+
+def _sig(i):
+    return (
+        f"intent_tags:intent:citations_required | prompt_hint:probe {i} | "
+        "tools: | env_keys:e2e"
+    )
+
+
+def test_match_batch_equals_sequential(tmp_path):
+    eng = _engine(tmp_path)
+    texts = [_sig(i) for i in range(8)]
+    ftypes = [None, "TIMEOUT", None, "HALLUCINATION_CITATION"] * 2
+    batched = eng.match_batch(texts, failure_types=ftypes)
+    for text, ftype, got in zip(texts, ftypes, batched):
+        want = eng.match(text, failure_type=ftype)
+        assert [(m.failure_id, m.version, round(m.score, 5)) for m in got] == [
+            (m.failure_id, m.version, round(m.score, 5)) for m in want
+        ]
+
+
+def test_match_lock_not_held_across_search(tmp_path):
+    """match() must not hold the engine lock while the kernel runs: a
+    concurrent upsert acquires the lock while a slow search is in
+    flight."""
+    eng = _engine(tmp_path)
+    entered = threading.Event()
+    release = threading.Event()
+    orig_search = eng.store.search
+
+    def slow_search(q, k, valid_n=None):
+        entered.set()
+        assert release.wait(timeout=10), "test deadlock"
+        return orig_search(q, k, valid_n=valid_n)
+
+    eng.store.search = slow_search
+    result = {}
+
+    def do_match():
+        result["m"] = eng.match(_sig(0))
+
+    t = threading.Thread(target=do_match)
+    t.start()
+    assert entered.wait(timeout=10)
+    # the search is blocked mid-flight; an upsert must still get the lock
+    got_lock = eng._lock.acquire(timeout=5)
+    assert got_lock, "engine lock held across the search kernel"
+    eng._lock.release()
+    eng.upsert_failure("TIMEOUT", _sig(99), {}, app_id="x")
+    release.set()
+    t.join(timeout=10)
+    assert result["m"] and result["m"][0].failure_id
+
+
+def test_micro_batcher_coalesces_concurrent_requests():
+    """N concurrent callers share fewer engine launches than requests."""
+
+    class SlowEngine:
+        def __init__(self):
+            self.batch_sizes = []
+
+        def match_batch(self, texts, ftypes=None, top_k=None):
+            self.batch_sizes.append(len(texts))
+            time.sleep(0.05)  # lets the queue accumulate the other callers
+            return [[("r", t)] for t in texts]
+
+    eng = SlowEngine()
+    batcher = MatchBatcher(eng)
+    n = 16
+    barrier = threading.Barrier(n)
+    results = [None] * n
+
+    def call(i):
+        barrier.wait()
+        results[i] = batcher.match(f"text-{i}")
+
+    threads = [threading.Thread(target=call, args=(i,)) for i in range(n)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=30)
+    assert all(r == [("r", f"text-{i}")] for i, r in enumerate(results))
+    assert sum(eng.batch_sizes) == n
+    assert len(eng.batch_sizes) < n, eng.batch_sizes  # coalescing happened
+    assert max(eng.batch_sizes) > 1
+    assert batcher.requests == n and batcher.batches == len(eng.batch_sizes)
+
+
+def test_micro_batcher_propagates_errors():
+    class FailingEngine:
+        def match_batch(self, texts, ftypes=None, top_k=None):
+            raise ValueError("boom")
+
+    batcher = MatchBatcher(FailingEngine())
+    import pytest
+
+    with pytest.raises(ValueError, match="boom"):
+        batcher.match("x")
+    # the batcher thread survives an error and serves the next request
+    class OkEngine:
+        def match_batch(self, texts, ftypes=None, top_k=None):
+            return [["ok"] for _ in texts]
+
+    batcher.engine = OkEngine()
+    assert batcher.match("y") == ["ok"]
+
+
+def test_service_match_through_batcher(tmp_path):
+    """/failures/match answers identically through the micro-batcher."""
+    from fastapi.testclient import TestClient
+
+    from kakveda_amd.services.gfkb_service import create_app
+
+    eng = _engine(tmp_path)
+    app = create_app(engine=eng)
+    assert app.state.batcher is not None
+    with TestClient(app) as client:
+        r = client.post(
+            "/failures/match",
+            json={"signature_text": _sig(3), "app_id": "app-0"},
+        )
+        assert r.status_code == 200
+        got = r.json()["matches"]
+    want = eng.match(_sig(3))
+    assert got and got[0]["failure_id"] == want[0].failure_id
+    assert abs(got[0]["score"] - want[0].score) < 1e-6
+    assert app.state.batcher.requests >= 1
