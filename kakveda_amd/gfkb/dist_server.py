@@ -148,7 +148,9 @@ class _CoordinatorStore:
     def append(self, rows: torch.Tensor) -> int:
         return self.coord.append(rows.to(self.dtype))
 
-    def search(self, queries: torch.Tensor, k: int):
+    def search(self, queries: torch.Tensor, k: int, valid_n=None):
+        # valid_n snapshots are a single-store liveness optimisation; the
+        # coordinator path is SPMD-collective and searches its live count
         return self.coord.search(queries.to(self.dtype), k)
 
     @property
