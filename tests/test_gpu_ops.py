@@ -279,3 +279,56 @@ def test_kernel_variants_match_default(variant):
         assert r.returncode == 0, r.stderr[-1500:]
         outs[ksel] = float(r.stdout.split("CSUM")[1].strip())
     assert abs(outs[None] - outs[variant]) < 1e-2, outs
+
+
+def test_segmented_store_gpu_matches_ref():
+    """Multi-segment HBM store: per-segment fused kernels + merge equal a
+    single-kernel search over the same rows."""
+    from kakveda_amd import ops
+    from kakveda_amd.gfkb.engine import EmbeddingStore
+
+    torch.manual_seed(9)
+    dim, n = 768, 50000
+    data = _rand_unit(n, dim, seed=60)
+    store = EmbeddingStore(dim, device="cuda:0", capacity=1024, segment_rows=16384)
+    for s in range(0, n, 7000):
+        store.append(data[s : min(s + 7000, n)])
+    assert store.count == n and store.n_segments > 1
+    q = _rand_unit(64, dim, seed=61)
+    scores, idx = store.search(q, 5)
+    ref_s, ref_i = ops.cosine_topk(q, data, 5)
+    torch.cuda.synchronize()
+    assert torch.allclose(scores, ref_s, atol=1e-3), (scores - ref_s).abs().max()
+    gathered = (q.float() @ data.float().t()).gather(1, idx)
+    assert torch.allclose(gathered, scores, atol=1e-3)
+
+
+def test_sidecar_gpu_restore(tmp_path):
+    """GPU engine restart restores bf16 rows bit-exactly from the packed
+    sidecar without re-encoding."""
+    from kakveda_amd.gfkb.engine import GfkbEngine
+
+    eng = GfkbEngine(data_dir=str(tmp_path), device="cuda", dim=768, hash_dim=16384)
+    sigs = [
+        f"intent_tags:intent:citations_required | prompt_hint:gpu restore {i} | "
+        "tools: | env_keys:e2e"
+        for i in range(32)
+    ]
+    for s in sigs:
+        eng.upsert_failure("HALLUCINATION_CITATION", s, {}, app_id="a")
+    rows = eng.store.row_range(0, 32).clone()
+    assert eng.sidecar.count() == 32
+
+    from kakveda_amd.encoder.model import TraceEncoder
+
+    calls = []
+    orig = TraceEncoder.encode_texts
+    TraceEncoder.encode_texts = lambda self, t: (calls.append(len(t)), orig(self, t))[1]
+    try:
+        eng2 = GfkbEngine(data_dir=str(tmp_path), device="cuda", dim=768, hash_dim=16384)
+    finally:
+        TraceEncoder.encode_texts = orig
+    assert eng2.store.count == 32 and not calls
+    assert torch.equal(eng2.store.row_range(0, 32), rows)
+    m = eng2.match(sigs[7])
+    assert m and m[0].score > 0.98
