@@ -152,14 +152,31 @@ def embedding_bag(table: torch.Tensor, idx: torch.Tensor, w: torch.Tensor) -> to
 # k-means (pattern detector)
 # ---------------------------------------------------------------------------
 
-def kmeans_assign(points: torch.Tensor, centroids: torch.Tensor) -> torch.Tensor:
-    """Nearest (max-cosine) centroid id per point -> int64 [N]."""
+def kmeans_assign_scored(
+    points: torch.Tensor, centroids: torch.Tensor
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(top-1 cosine f32 [N], nearest centroid id i64 [N]) per point.
+
+    GPU, C<=64: the dedicated LDS-resident-centroid kernel (point stream
+    straight from HBM to MFMA, no per-window barriers — built for the
+    pattern detector's N-huge/C-tiny shape). GPU, C>64: the general fused
+    cosine_topk with k=1. CPU: eager matmul+argmax.
+    """
     if points.device.type == "cuda":
         ext = _require_ext()
-        _, idx = ext.cosine_topk(points, centroids, 1, centroids.shape[0])
-        return idx[:, 0]
+        if centroids.shape[0] <= 64:
+            scores, idx = ext.kmeans_assign(points, centroids)
+            return scores, idx.long()
+        scores, idx = ext.cosine_topk(points, centroids, 1, centroids.shape[0])
+        return scores[:, 0], idx[:, 0]
     sims = points.float() @ centroids.float().t()
-    return sims.argmax(dim=1)
+    top = sims.max(dim=1)
+    return top.values, top.indices
+
+
+def kmeans_assign(points: torch.Tensor, centroids: torch.Tensor) -> torch.Tensor:
+    """Nearest (max-cosine) centroid id per point -> int64 [N]."""
+    return kmeans_assign_scored(points, centroids)[1]
 
 
 def kmeans_update(
@@ -170,7 +187,7 @@ def kmeans_update(
     GPU: LDS-partial segmented-reduction HIP kernel (points read once per
     64-dim tile). CPU fallback: index_add_.
     """
-    if points.device.type == "cuda" and n_clusters <= 256:
+    if points.device.type == "cuda" and n_clusters <= 512:
         ext = _require_ext()
         return ext.kmeans_update(
             points.to(torch.bfloat16), assign.to(torch.int32), int(n_clusters)

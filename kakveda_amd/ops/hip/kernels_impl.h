@@ -1503,5 +1503,133 @@ __global__ __launch_bounds__(THREADS) void embedding_bag_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// k-means assignment: N points x C centroids (C <= 64), argmax per point.
+//
+// The general cosine_topk path is built for huge corpora: at the pattern-
+// detector shape (N=10M points as "queries", C=64 centroids as the
+// "corpus") its 12-window staged pipeline runs with one 64-valid-column
+// tile, nothing to overlap, and a per-block B-restage — measured ~440 GB/s
+// effective vs the ~8 TB/s point-stream bound (ROUND2.md). This kernel is
+// shaped for that case instead:
+//   - centroids are staged into LDS ONCE per block (row-padded pitch so the
+//     16-lane fragment reads spread across all 64 banks) and stay resident
+//     for every point tile the block processes;
+//   - point MFMA A-fragments stream straight from HBM into registers
+//     (16 B/lane, sector-coalesced) through a 1-window register prefetch —
+//     no LDS staging, no per-window barriers at all;
+//   - 8 waves x 32 rows = 256 points per tile, grid-strided so one launch
+//     covers any N; the argmax epilogue is pure registers + shfl.
+// ---------------------------------------------------------------------------
+constexpr int ASSIGN_THREADS = 512;  // 8 waves
+
+__global__ __launch_bounds__(ASSIGN_THREADS, 1) void kmeans_assign_kernel(
+    const bf16_t* __restrict__ P, const bf16_t* __restrict__ Cc,
+    float* __restrict__ out_score, int* __restrict__ out_idx,
+    int N, int D, int C) {
+  extern __shared__ char cmem[];  // 64 centroid rows, padded pitch
+  const int pitch = D * 2 + 16;   // +16 B: successive rows shift 4 banks
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int g = lane >> 4, cl = lane & 15;
+
+  // stage centroids once; rows >= C zero-filled so their MFMAs stay finite
+  for (int s = tid; s < 64 * (D / 8); s += ASSIGN_THREADS) {
+    const int c = s / (D / 8);
+    const int so = s % (D / 8);
+    bf16x8 v = {};
+    if (c < C)
+      v = *(const bf16x8*)((const char*)Cc + ((size_t)c * D + (size_t)so * 8) * 2);
+    *(bf16x8*)(cmem + (size_t)c * pitch + so * 16) = v;
+  }
+  __syncthreads();
+
+  const long prow = (long)D * 2;
+  const int nkt = D / 64;
+  for (long tile0 = (long)blockIdx.x * 256; tile0 < N;
+       tile0 += (long)gridDim.x * 256) {
+    const long r0 = tile0 + (long)wid * 32;  // this wave's 32 rows
+    f32x4 acc[2][4];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int n = 0; n < 4; ++n) acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    // A fragments direct global->register, 1 window ahead (tail rows clamp
+    // to the last point; their results are masked at the write below)
+    bf16x8 a[2][2][2];  // [buf][kk][m]
+    auto lda = [&](int kt, int kk, int m) -> bf16x8 {
+      long r = r0 + m * 16 + cl;
+      if (r >= N) r = N - 1;
+      return *(const bf16x8*)((const char*)P + r * prow +
+                              (size_t)(kt * 8 + kk * 4 + g) * 16);
+    };
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int m = 0; m < 2; ++m) a[0][kk][m] = lda(0, kk, m);
+    int buf = 0;
+    for (int kt = 0; kt < nkt; ++kt) {
+      if (kt + 1 < nkt) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+          for (int m = 0; m < 2; ++m) a[buf ^ 1][kk][m] = lda(kt + 1, kk, m);
+      }
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x8 bfrag[4];
+        const int slot = kt * 8 + kk * 4 + g;
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          bfrag[n] =
+              *(const bf16x8*)(cmem + (size_t)(n * 16 + cl) * pitch + slot * 16);
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a[buf][kk][m], bfrag[n], acc[m][n], 0, 0, 0);
+      }
+      buf ^= 1;
+    }
+
+    // argmax per row over 64 cols (cols >= C masked; ties -> lowest col,
+    // matching torch.argmax / the CPU reference)
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        float best = NEG_INF;
+        int bcol = 0x7fffffff;
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int col = n * 16 + cl;
+          const float v = acc[m][n][reg];
+          if (col < C && (v > best || (v == best && col < bcol))) {
+            best = v;
+            bcol = col;
+          }
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          const float ov = __shfl_xor(best, off, 64);
+          const int oc = __shfl_xor(bcol, off, 64);
+          if (ov > best || (ov == best && oc < bcol)) {
+            best = ov;
+            bcol = oc;
+          }
+        }
+        const long row = r0 + m * 16 + g * 4 + reg;
+        if (cl == 0 && row < N) {
+          out_score[row] = best;
+          out_idx[row] = bcol;
+        }
+      }
+    }
+  }
+}
+
 }  // namespace kakveda
 

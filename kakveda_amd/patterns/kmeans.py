@@ -60,11 +60,10 @@ class StreamingKMeans:
         """
         pts = points.to(self.device)
         if self.device.type == "cuda":
-            scores, idx = ops.cosine_topk(
-                pts.to(torch.bfloat16), self._centroids_matcher(), 1, self.k
+            scores, assign = ops.kmeans_assign_scored(
+                pts.to(torch.bfloat16), self._centroids_matcher()
             )
-            assign = idx[:, 0]
-            mean_cos = scores[:, 0].mean()
+            mean_cos = scores.mean()
         else:
             sims = pts.float() @ self.centroids.t()
             top = sims.max(dim=1)

@@ -332,3 +332,42 @@ def test_sidecar_gpu_restore(tmp_path):
     assert torch.equal(eng2.store.row_range(0, 32), rows)
     m = eng2.match(sigs[7])
     assert m and m[0].score > 0.98
+
+
+@pytest.mark.parametrize("C,N", [(64, 10007), (5, 300), (64, 256), (33, 70000)])
+def test_kmeans_assign_kernel_vs_torch(C, N):
+    """Dedicated LDS-resident-centroid assignment kernel (C<=64) matches
+    the fp32 torch argmax on the same bf16 inputs."""
+    from kakveda_amd import ops
+
+    D = 768
+    pts = _rand_unit(N, D, seed=70 + C)
+    cents = _rand_unit(C, D, seed=71 + C)
+    scores, idx = ops.kmeans_assign_scored(pts, cents)
+    torch.cuda.synchronize()
+    assert scores.shape == (N,) and idx.shape == (N,)
+    sims = pts.float() @ cents.float().t()
+    ref_v, ref_i = sims.max(dim=1)
+    # scores match the fp32 reference within bf16-accumulation tolerance
+    assert torch.allclose(scores, ref_v, atol=2e-2, rtol=1e-2), (
+        (scores - ref_v).abs().max().item()
+    )
+    # each claimed centroid really achieves the claimed score
+    gathered = sims.gather(1, idx.unsqueeze(1))[:, 0]
+    assert torch.allclose(gathered, scores, atol=1e-4)
+    # and the claimed centroid is a true argmax (scores equal at the max)
+    assert torch.allclose(gathered, ref_v, atol=1e-4)
+
+
+def test_kmeans_assign_tie_breaks_to_first():
+    """Duplicate centroids: ties resolve to the lowest centroid id,
+    matching torch.argmax/CPU-reference first-occurrence semantics."""
+    from kakveda_amd import ops
+
+    D = 768
+    cents = _rand_unit(8, D, seed=80)
+    cents = torch.cat([cents, cents[:4]])  # ids 8..11 duplicate 0..3
+    pts = cents[torch.randint(0, 12, (500,), generator=torch.Generator().manual_seed(81))]
+    _, idx = ops.kmeans_assign_scored(pts.contiguous(), cents.contiguous())
+    torch.cuda.synchronize()
+    assert (idx < 8).all(), idx.max()
