@@ -15,7 +15,7 @@ import json
 import os
 import secrets
 import time
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List, Optional  # noqa: F401
 
 PBKDF2_ITERS = int(os.environ.get("KAKVEDA_PBKDF2_ITERS", "260000"))
 JWT_TTL_MIN = 720
@@ -107,18 +107,57 @@ def require_any(payload: Optional[Dict[str, Any]], roles: List[str]) -> bool:
     return bool(payload) and any(r in (payload.get("roles") or []) for r in roles)
 
 
-# -- revocation + rate limiting (in-memory; parity with the reference's
-#    Redis-or-memory fallback, shared/redis_helpers.py:26-84) ---------------
+# -- revocation + rate limiting (Redis-backed with in-memory fallback;
+#    parity with the reference's shared/redis_helpers.py:26-84:
+#    SET-with-TTL revocation, fixed-window INCR+EXPIRE rate limiting) ------
+
+
+def _redis_client(redis_url: Optional[str], injected: Any = None) -> Any:
+    """A live Redis client, or None (missing package / unreachable server
+    -> silent in-memory fallback, like the reference)."""
+    if injected is not None:
+        return injected
+    if not redis_url:
+        return None
+    try:
+        import redis  # optional dependency
+
+        client = redis.Redis.from_url(redis_url)
+        client.ping()
+        return client
+    except Exception:
+        return None
 
 
 class RevocationStore:
-    def __init__(self):
+    """Revoked JWT jti set. Redis: SET key with TTL (shared across
+    replicas); fallback: per-process dict with expiry."""
+
+    def __init__(self, redis_url: Optional[str] = None, prefix: str = "kv:revoked:",
+                 client: Any = None):
+        self.prefix = prefix
+        self._redis = _redis_client(redis_url, client)
         self._revoked: Dict[str, float] = {}
 
+    @property
+    def backend(self) -> str:
+        return "redis" if self._redis is not None else "memory"
+
     def revoke(self, jti: str, ttl_sec: float = JWT_TTL_MIN * 60):
+        if self._redis is not None:
+            try:
+                self._redis.set(self.prefix + jti, "1", ex=max(1, int(ttl_sec)))
+                return
+            except Exception:
+                self._redis = None  # degrade to memory mid-flight
         self._revoked[jti] = time.time() + ttl_sec
 
     def is_revoked(self, jti: str) -> bool:
+        if self._redis is not None:
+            try:
+                return self._redis.get(self.prefix + jti) is not None
+            except Exception:
+                self._redis = None
         exp = self._revoked.get(jti)
         if exp is None:
             return False
@@ -129,15 +168,33 @@ class RevocationStore:
 
 
 class RateLimiter:
-    """Fixed-window counter per key."""
+    """Fixed-window counter per key. Redis: INCR + EXPIRE on a
+    per-window key (shared across replicas); fallback: per-process."""
 
-    def __init__(self, limit: int = 30, window_sec: float = 60.0):
+    def __init__(self, limit: int = 30, window_sec: float = 60.0,
+                 redis_url: Optional[str] = None, prefix: str = "kv:rl:",
+                 client: Any = None):
         self.limit = limit
         self.window = window_sec
+        self.prefix = prefix
+        self._redis = _redis_client(redis_url, client)
         self._counts: Dict[str, tuple[int, float]] = {}
+
+    @property
+    def backend(self) -> str:
+        return "redis" if self._redis is not None else "memory"
 
     def allow(self, key: str) -> bool:
         now = time.time()
+        if self._redis is not None:
+            try:
+                wkey = f"{self.prefix}{key}:{int(now // self.window)}"
+                n = int(self._redis.incr(wkey))
+                if n == 1:
+                    self._redis.expire(wkey, int(self.window) + 1)
+                return n <= self.limit
+            except Exception:
+                self._redis = None
         count, start = self._counts.get(key, (0, now))
         if now - start > self.window:
             count, start = 0, now

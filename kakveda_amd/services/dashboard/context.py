@@ -45,14 +45,21 @@ class DashboardContext:
     urls: Dict[str, str]
     data_dir: str
     jwt_secret: str = ""
-    revocation: RevocationStore = field(default_factory=RevocationStore)
-    auth_limiter: RateLimiter = field(default_factory=lambda: RateLimiter(20, 60))
+    revocation: Optional[RevocationStore] = None
+    auth_limiter: Optional[RateLimiter] = None
     log: Any = None
 
     def __post_init__(self):
         rc = get_runtime_config()
         if not self.jwt_secret:
             self.jwt_secret = rc.jwt_secret
+        # Redis-backed revocation/rate-limit when KAKVEDA_REDIS_URL is set
+        # and reachable; in-memory fallback otherwise (reference
+        # shared/redis_helpers.py:26-84 semantics)
+        if self.revocation is None:
+            self.revocation = RevocationStore(redis_url=rc.redis_url)
+        if self.auth_limiter is None:
+            self.auth_limiter = RateLimiter(20, 60, redis_url=rc.redis_url)
         if rc.env == "prod" and self.jwt_secret == "kakveda-dev-secret":
             # production guardrail (reference dashboard/app.py:1266-1269)
             raise RuntimeError(

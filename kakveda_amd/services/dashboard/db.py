@@ -273,16 +273,66 @@ class EvaluationResult(Base):
     detail_json: Mapped[str] = mapped_column(Text, default="{}")
 
 
-def make_engine(db_path: str):
-    eng = create_engine(
-        f"sqlite:///{db_path}", connect_args={"check_same_thread": False}
-    )
-    return eng
+def make_engine(db_path: Optional[str] = None, url: Optional[str] = None):
+    """Engine from an explicit URL, the DATABASE_URL env (reference
+    docker-compose.prod.yml runs the dashboard on postgres this way), or
+    a local SQLite file path."""
+    url = url or os.environ.get("DATABASE_URL") or ""
+    if not url:
+        url = f"sqlite:///{db_path}"
+    kwargs = {}
+    if url.startswith("sqlite"):
+        kwargs["connect_args"] = {"check_same_thread": False}
+    else:
+        kwargs["pool_pre_ping"] = True
+    return create_engine(url, **kwargs)
 
 
-def init_db(db_path: str):
-    """Create (or complete) the schema; returns a session factory."""
-    os.makedirs(os.path.dirname(db_path) or ".", exist_ok=True)
-    eng = make_engine(db_path)
+def _default_sql(col) -> str:
+    """DEFAULT clause for ALTER TABLE ADD COLUMN, from the model default."""
+    d = getattr(col.default, "arg", None)
+    if d is None or callable(d):
+        return ""
+    if isinstance(d, bool):
+        return f" DEFAULT {1 if d else 0}"
+    if isinstance(d, (int, float)):
+        return f" DEFAULT {d}"
+    return " DEFAULT '{}'".format(str(d).replace("'", "''"))
+
+
+def migrate_db(eng) -> list:
+    """Idempotent startup migrations (reference services/dashboard/db.py:
+    368-644 runs hand-rolled ALTERs on every boot): add any model column
+    missing from an existing table via ALTER TABLE ADD COLUMN (with the
+    model's scalar default), then create any missing tables. Safe to run
+    on every startup; returns the DDL statements applied."""
+    from sqlalchemy import inspect, text
+
+    insp = inspect(eng)
+    applied = []
+    existing = set(insp.get_table_names())
+    with eng.begin() as conn:
+        for table in Base.metadata.sorted_tables:
+            if table.name not in existing:
+                continue  # create_all below creates it whole
+            have = {c["name"] for c in insp.get_columns(table.name)}
+            for col in table.columns:
+                if col.name in have:
+                    continue
+                ddl = (
+                    f"ALTER TABLE {table.name} ADD COLUMN {col.name} "
+                    f"{col.type.compile(eng.dialect)}{_default_sql(col)}"
+                )
+                conn.execute(text(ddl))
+                applied.append(ddl)
     Base.metadata.create_all(eng)
+    return applied
+
+
+def init_db(db_path: Optional[str] = None, url: Optional[str] = None):
+    """Create (or migrate-forward) the schema; returns a session factory."""
+    if db_path and not (url or os.environ.get("DATABASE_URL")):
+        os.makedirs(os.path.dirname(db_path) or ".", exist_ok=True)
+    eng = make_engine(db_path, url=url)
+    migrate_db(eng)
     return sessionmaker(bind=eng, expire_on_commit=False)
