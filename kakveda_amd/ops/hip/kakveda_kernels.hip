@@ -253,7 +253,7 @@ std::tuple<torch::Tensor, torch::Tensor> kmeans_assign(
   TORCH_CHECK(centroids.size(1) == D, "dim mismatch");
   TORCH_CHECK(D % 64 == 0, "D must be a multiple of 64");
   TORCH_CHECK(C >= 1 && C <= 64, "kmeans_assign kernel supports C<=64");
-  const size_t lds = 64 * ((size_t)D * 2 + 16);
+  const size_t lds = 64 * (size_t)assign_pitch(D);
   TORCH_CHECK(lds <= 160 * 1024, "D too large for LDS-resident centroids");
   auto opts_f = torch::TensorOptions().dtype(torch::kFloat32).device(points.device());
   auto opts_i = torch::TensorOptions().dtype(torch::kInt32).device(points.device());

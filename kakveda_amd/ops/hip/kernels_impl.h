@@ -1523,12 +1523,21 @@ __global__ __launch_bounds__(THREADS) void embedding_bag_kernel(
 // ---------------------------------------------------------------------------
 constexpr int ASSIGN_THREADS = 512;  // 8 waves
 
+// Row pitch for the centroid LDS image: pad so pitch % 256 == 64, which
+// makes the 64 lanes of a fragment read (16 rows x 4 slot-groups of 16 B)
+// land on 64 distinct banks — the minimum 4 conflict-free phases for a
+// 1 KiB ds_read_b128 wave. (A 16 B pad measured as a 4x conflict: lanes
+// with equal cl+g collided.)
+constexpr __host__ __device__ int assign_pitch(int D) {
+  return D * 2 + ((64 - (D * 2) % 256) + 256) % 256;
+}
+
 __global__ __launch_bounds__(ASSIGN_THREADS, 1) void kmeans_assign_kernel(
     const bf16_t* __restrict__ P, const bf16_t* __restrict__ Cc,
     float* __restrict__ out_score, int* __restrict__ out_idx,
     int N, int D, int C) {
   extern __shared__ char cmem[];  // 64 centroid rows, padded pitch
-  const int pitch = D * 2 + 16;   // +16 B: successive rows shift 4 banks
+  const int pitch = assign_pitch(D);
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -1556,27 +1565,29 @@ __global__ __launch_bounds__(ASSIGN_THREADS, 1) void kmeans_assign_kernel(
 #pragma unroll
       for (int n = 0; n < 4; ++n) acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    // A fragments direct global->register, 1 window ahead (tail rows clamp
-    // to the last point; their results are masked at the write below)
-    bf16x8 a[2][2][2];  // [buf][kk][m]
-    auto lda = [&](int kt, int kk, int m) -> bf16x8 {
-      long r = r0 + m * 16 + cl;
-      if (r >= N) r = N - 1;
-      return *(const bf16x8*)((const char*)P + r * prow +
-                              (size_t)(kt * 8 + kk * 4 + g) * 16);
+    // A fragments direct global->register. Straight-line, unroll-by-3
+    // window loop with THREE named prefetch buffers and 2 windows always
+    // in flight: a branchy 1-deep version compiled to a `s_waitcnt
+    // vmcnt(0)` before every MFMA phase (the conditional prefetch block
+    // defeats counted waits) and measured 153 GB/s; straight-line code
+    // lets the compiler emit counted vmcnt waits so the point stream
+    // pipelines. Tail rows clamp to the last point (results masked at
+    // the write below); past-the-end windows clamp to the last window
+    // (redundant re-reads, still straight-line).
+    bf16x8 a0[2][2], a1[2][2], a2[2][2];  // [kk][m]
+    auto lda = [&](bf16x8 (&dst)[2][2], int kt) {
+      if (kt >= nkt) kt = nkt - 1;
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int m = 0; m < 2; ++m) {
+          long r = r0 + m * 16 + cl;
+          if (r >= N) r = N - 1;
+          dst[kk][m] = *(const bf16x8*)((const char*)P + r * prow +
+                                        (size_t)(kt * 8 + kk * 4 + g) * 16);
+        }
     };
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk)
-#pragma unroll
-      for (int m = 0; m < 2; ++m) a[0][kk][m] = lda(0, kk, m);
-    int buf = 0;
-    for (int kt = 0; kt < nkt; ++kt) {
-      if (kt + 1 < nkt) {
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk)
-#pragma unroll
-          for (int m = 0; m < 2; ++m) a[buf ^ 1][kk][m] = lda(kt + 1, kk, m);
-      }
+    auto mfma_win = [&](bf16x8 (&src)[2][2], int kt) {
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
         bf16x8 bfrag[4];
@@ -1590,10 +1601,22 @@ __global__ __launch_bounds__(ASSIGN_THREADS, 1) void kmeans_assign_kernel(
 #pragma unroll
           for (int n = 0; n < 4; ++n)
             acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a[buf][kk][m], bfrag[n], acc[m][n], 0, 0, 0);
+                src[kk][m], bfrag[n], acc[m][n], 0, 0, 0);
       }
-      buf ^= 1;
+    };
+    lda(a0, 0);
+    lda(a1, 1);
+    int kt = 0;
+    for (; kt + 2 < nkt; kt += 3) {
+      lda(a2, kt + 2);
+      mfma_win(a0, kt);
+      lda(a0, kt + 3);
+      mfma_win(a1, kt + 1);
+      lda(a1, kt + 4);
+      mfma_win(a2, kt + 2);
     }
+    if (kt < nkt) mfma_win(a0, kt);
+    if (kt + 1 < nkt) mfma_win(a1, kt + 1);
 
     // argmax per row over 64 cols (cols >= C masked; ties -> lowest col,
     // matching torch.argmax / the CPU reference)
