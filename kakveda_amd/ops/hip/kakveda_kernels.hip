@@ -288,8 +288,12 @@ std::tuple<torch::Tensor, torch::Tensor> kmeans_update(
   const int C = (int)n_clusters;
   TORCH_CHECK(D % 64 == 0, "D must be a multiple of 64");
   TORCH_CHECK(C >= 1 && C <= 512, "n_clusters must be in [1,512]");
-  if (C > 256) {
-    // the [C][65] f32 LDS partial exceeds the 64 KiB default dynamic cap
+  // replicated partials for C<=128 (4x[C][64]) or single-replica above;
+  // either can exceed the 64 KiB default dynamic cap
+  const int repl = (C <= 128) ? 4 : 1;
+  const size_t lds_upd = ((size_t)repl * C * 64 + C) * 4;
+  TORCH_CHECK(lds_upd <= 160 * 1024, "n_clusters LDS overflow");
+  if (lds_upd > 64 * 1024) {
     static bool attr_set = false;
     if (!attr_set) {
       (void)hipFuncSetAttribute((const void*)kmeans_update_kernel,
@@ -308,7 +312,7 @@ std::tuple<torch::Tensor, torch::Tensor> kmeans_update(
   chunks = (N + ppc - 1) / ppc;
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(kmeans_update_kernel, dim3(dim_tiles, chunks), dim3(256),
-                     C * 65 * 4, stream.stream(),
+                     lds_upd, stream.stream(),
                      (const bf16_t*)points.data_ptr(), assign.data_ptr<int>(),
                      sums.data_ptr<float>(), counts.data_ptr<float>(), N, D, C,
                      ppc);

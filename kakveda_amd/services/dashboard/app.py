@@ -189,6 +189,25 @@ def create_app(
         )
         return response
 
+    # -- static assets (theme css, logo, progressive-enhancement js) ---------
+
+    from kakveda_amd.services.dashboard.static import APP_JS, LOGO_SVG, STYLE_CSS
+
+    @app.get("/static/style.css")
+    async def style_css():
+        return Response(STYLE_CSS, media_type="text/css",
+                        headers={"Cache-Control": "public, max-age=3600"})
+
+    @app.get("/static/logo.svg")
+    async def logo_svg():
+        return Response(LOGO_SVG, media_type="image/svg+xml",
+                        headers={"Cache-Control": "public, max-age=3600"})
+
+    @app.get("/static/app.js")
+    async def app_js():
+        return Response(APP_JS, media_type="application/javascript",
+                        headers={"Cache-Control": "public, max-age=3600"})
+
     # -- auth ----------------------------------------------------------------
 
     @app.get("/login", response_class=HTMLResponse)

@@ -10,44 +10,77 @@ from typing import Any, Dict
 
 from jinja2 import DictLoader, Environment, select_autoescape
 
-_BASE = """<!doctype html><html><head><meta charset="utf-8">
-<title>kakveda-amd</title>
-<style>
-body{font-family:system-ui,sans-serif;margin:2rem;background:#0d1117;color:#e6edf3}
-a{color:#58a6ff} table{border-collapse:collapse;width:100%}
-td,th{border-bottom:1px solid #30363d;padding:.4rem .6rem;text-align:left;font-size:.9rem}
-.nav a{margin-right:1rem} .card{background:#161b22;border:1px solid #30363d;
-border-radius:8px;padding:1rem;margin:1rem 0}
-.bar{height:10px;background:#1f6feb;border-radius:3px}
-.badge{padding:.1rem .5rem;border-radius:1rem;background:#21262d;font-size:.8rem}
-input,textarea,select,button{background:#21262d;color:#e6edf3;border:1px solid #30363d;
-border-radius:6px;padding:.4rem .6rem;margin:.2rem 0}
-button{cursor:pointer;background:#238636}
-</style></head><body>
-<div class="nav"><a href="/">Dashboard</a><a href="/warnings">Warnings</a>
-<a href="/runs">Runs</a><a href="/playground">Playground</a>
-<a href="/agents">Agents</a><a href="/datasets">Datasets</a>
-<a href="/prompts">Prompts</a><a href="/experiments">Experiments</a>
-<a href="/health">Health</a><a href="/logout">Logout</a></div>
+from kakveda_amd.services.dashboard.static import LOGO_SVG
+
+_NAV = [
+    ("/", "Dashboard"),
+    ("/warnings", "Warnings"),
+    ("/scenarios", "Scenarios"),
+    ("/runs", "Runs"),
+    ("/playground", "Playground"),
+    ("/datasets", "Datasets"),
+    ("/evals", "Evals"),
+    ("/prompts", "Prompts"),
+    ("/experiments", "Experiments"),
+    ("/agents", "Agents"),
+    ("/projects", "Projects"),
+    ("/health", "Health"),
+]
+
+_BASE = (
+    """<!doctype html><html><head><meta charset="utf-8">
+<meta name="viewport" content="width=device-width,initial-scale=1">
+<title>{% block title %}kakveda-amd{% endblock %}</title>
+<link rel="stylesheet" href="/static/style.css">
+<link rel="icon" type="image/svg+xml" href="/static/logo.svg">
+<script src="/static/app.js" defer></script>
+</head><body>
+<header class="topbar">
+<a class="brand" href="/">"""
+    + LOGO_SVG
+    + """<span>kakveda<span class="amd">-amd</span></span></a>
+<nav class="main">
+{% for href, label in nav %}<a href="{{ href }}"{% if active == href %} class="active"{% endif %}>{{ label }}</a>{% endfor %}
+</nav>
+<div class="spacer"></div>
+{% if user %}<span class="who">{{ user }}</span>
+<nav class="main"><a href="/logout">Logout</a></nav>{% endif %}
+</header>
+<main>
 {% block content %}{% endblock %}
+</main>
+<footer>kakveda-amd — MI355X-native failure intelligence ·
+<a href="/healthz">healthz</a> · <a href="/admin/audit_page">audit</a></footer>
 </body></html>"""
+)
 
 _TEMPLATES = {
     "base.html": _BASE,
     "login.html": """{% extends "base.html" %}{% block content %}
-<div class="card"><h2>Sign in</h2>
+<div class="card auth"><div class="brand">""" + LOGO_SVG + """</div>
+<h2>Sign in</h2>
 <form method="post" action="/login">
-<input name="email" placeholder="email" value="admin@kakveda.local"><br>
-<input name="password" type="password" placeholder="password"><br>
+<input name="email" placeholder="email" value="admin@kakveda.local">
+<input name="password" type="password" placeholder="password">
 <button type="submit">Login</button></form>
-<p>Demo users: admin/operator/viewer/demo @kakveda.local</p></div>
-{% endblock %}""",
+<p class="dim">Demo users: admin/operator/viewer/demo @kakveda.local</p>
+<p><a href="/register">Create account</a> · <a href="/forgot">Forgot password</a></p>
+</div>{% endblock %}""",
     "home.html": """{% extends "base.html" %}{% block content %}
 <h1>Failure Intelligence</h1>
-<p>Signed in as {{ user }} ({{ roles|join(", ") }})</p>
+<p class="dim">Signed in as {{ user }} ({{ roles|join(", ") }})</p>
+<div class="tiles">
+<div class="tile"><div class="n">{{ failures|length }}</div><div class="l">recent failures</div></div>
+<div class="tile"><div class="n">{{ patterns|length }}</div><div class="l">patterns</div></div>
+<div class="tile"><div class="n">{{ warnings|length }}</div><div class="l">latest warnings</div></div>
+<div class="tile"><div class="n">{{ runs|length }}</div><div class="l">recent runs</div></div>
+</div>
+<div class="grid2">
 <div class="card"><h3>Recent failures (GFKB)</h3><table>
-<tr><th>id</th><th>v</th><th>type</th><th>apps</th><th>occurrences</th></tr>
-{% for f in failures %}<tr><td>{{ f.failure_id }}</td><td>{{ f.version }}</td>
+<tr><th>id</th><th>v</th><th>type</th><th>apps</th><th>occ</th></tr>
+{% for f in failures %}<tr>
+<td><a href="/failure/{{ f.failure_id }}">{{ f.failure_id }}</a></td>
+<td>{{ f.version }}</td>
 <td>{{ f.failure_type }}</td><td>{{ f.affected_apps|join(", ") }}</td>
 <td>{{ f.occurrences }}</td></tr>{% endfor %}</table></div>
 <div class="card"><h3>Patterns</h3><table>
@@ -55,13 +88,14 @@ _TEMPLATES = {
 {% for p in patterns %}<tr><td>{{ p.pattern_id }}</td><td>{{ p.name }}</td>
 <td>{{ p.affected_apps|join(", ") }}</td><td>{{ p.failure_ids|length }}</td></tr>
 {% endfor %}</table></div>
+</div>
 <div class="card"><h3>Latest warnings</h3><table>
 <tr><th>ts</th><th>app</th><th>action</th><th>confidence</th></tr>
 {% for w in warnings %}<tr><td>{{ w.ts }}</td><td>{{ w.app_id }}</td>
-<td><span class="badge">{{ w.action }}</span></td>
+<td><span class="badge {{ w.action }}">{{ w.action }}</span></td>
 <td>{{ "%.2f"|format(w.confidence) }}</td></tr>{% endfor %}</table></div>
 <div class="card"><h3>Run a scenario</h3>
-<form method="post" action="/scenarios/run">
+<form class="inline" method="post" action="/scenarios/run">
 <input name="app_id" value="app-A">
 <input name="prompt" size="60"
  value="Summarize this and include references even if none are provided.">
@@ -69,11 +103,34 @@ _TEMPLATES = {
 {% endblock %}""",
     "warnings.html": """{% extends "base.html" %}{% block content %}
 <h1>Warnings</h1>
-<div class="card"><h3>Last {{ analytics.days }} days: {{ analytics.total }} warnings,
-est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
-<table><tr><th>day</th><th>count</th></tr>
-{% for day, n in analytics.daily.items() %}<tr><td>{{ day }}</td><td>{{ n }}</td></tr>{% endfor %}
-</table></div>
+<div class="tiles">
+<div class="tile"><div class="n">{{ analytics.total }}</div>
+<div class="l">last {{ analytics.days }} days</div></div>
+<div class="tile"><div class="n">{{ analytics.per_app|length }}</div>
+<div class="l">apps affected</div></div>
+<div class="tile"><div class="n">{{ analytics.per_pattern|length }}</div>
+<div class="l">patterns hit</div></div>
+<div class="tile"><div class="n">{{ analytics.est_cost_impact_usd_micro }}</div>
+<div class="l">est. cost µUSD</div></div>
+</div>
+{% if analytics.daily %}
+{% set peak = analytics.daily.values()|max %}
+<div class="card"><h3>Daily warnings</h3>
+<div class="chart">
+{% for day, n in analytics.daily.items() %}
+<div class="col"><div class="bar" style="height:{{ (100 * n / peak)|round }}%"
+ title="{{ day }}: {{ n }}"></div><span class="lab">{{ day[5:] }}</span></div>
+{% endfor %}
+</div></div>
+{% endif %}
+<div class="grid2">
+<div class="card"><h3>Per app</h3><table><tr><th>app</th><th>count</th></tr>
+{% for app, n in analytics.per_app.items() %}
+<tr><td>{{ app }}</td><td>{{ n }}</td></tr>{% endfor %}</table></div>
+<div class="card"><h3>Per pattern</h3><table><tr><th>pattern</th><th>count</th></tr>
+{% for p, n in analytics.per_pattern.items() %}
+<tr><td>{{ p }}</td><td>{{ n }}</td></tr>{% endfor %}</table></div>
+</div>
 <div class="card"><table>
 <tr><th>ts</th><th>app</th><th>action</th><th>conf</th><th>pattern</th><th>message</th></tr>
 {% for w in warnings %}<tr id="w-{{ w.id }}"><td>{{ w.ts }}</td><td>{{ w.app_id }}</td>
@@ -106,9 +163,18 @@ est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
 </div>{% endfor %}</div>{% endblock %}""",
     "playground.html": """{% extends "base.html" %}{% block content %}
 <h1>Playground</h1>
-<div class="card"><p>POST /api/playground/run with {"prompt", "model",
+<div class="card">
+<form id="pg-form">
+<select name="model">
+{% for m in models %}<option value="{{ m }}">{{ m }}</option>{% endfor %}
+</select><br>
+<textarea name="prompt" rows="5" style="width:100%"
+ placeholder="Ask something… the deterministic stub answers with citations when no model backend is reachable."></textarea><br>
+<button type="submit">Run</button>
+</form>
+<h3>Output</h3><pre id="pg-out" class="dim">—</pre>
+<p class="dim">API: POST /api/playground/run with {"prompt", "model",
 "agent_id", "prompt_version_id", "experiment"}.</p>
-<p>Models: {% for m in models %}<span class="badge">{{ m }}</span> {% endfor %}</p>
 </div>{% endblock %}""",
     "agents.html": """{% extends "base.html" %}{% block content %}
 <h1>Agent registry</h1>
@@ -165,7 +231,15 @@ est. cost impact {{ analytics.est_cost_impact_usd_micro }} µUSD</h3>
     "health.html": """{% extends "base.html" %}{% block content %}
 <h1>App health</h1>
 {% for app_id, points in apps.items() %}
-<div class="card"><h3>{{ app_id }}</h3><table>
+{% set latest = points[-1] %}
+<div class="card"><h3>{{ app_id }}</h3>
+<div class="score">
+<b>{{ "%.1f"|format(latest.score) }}</b>
+<div class="track"><div class="bar
+ {% if latest.score >= 80 %}good{% elif latest.score >= 50 %}mid{% else %}low{% endif %}"
+ style="width:{{ [latest.score, 0]|max }}%"></div></div>
+</div>
+<table>
 <tr><th>ts</th><th>score</th><th>failure rate</th><th>penalty</th></tr>
 {% for p in points %}<tr><td>{{ p.ts }}</td><td>{{ "%.1f"|format(p.score) }}</td>
 <td>{{ "%.2f"|format(p.failure_rate) }}</td><td>{{ p.recurrent_penalty }}</td></tr>
@@ -220,20 +294,20 @@ POST /health/test.</div>{% endfor %}{% endblock %}""",
 <td>{{ r.warn_action }}</td><td>{{ "%.2f"|format(r.warn_confidence) }}</td></tr>
 {% endfor %}</table></div>{% endblock %}""",
     "register.html": """{% extends "base.html" %}{% block content %}
-<div class="card"><h2>Create account</h2>
+<div class="card auth"><h2>Create account</h2>
 <form method="post" action="/register">
 <input name="email" placeholder="email"><br>
 <input name="password" type="password" placeholder="password (8+ chars)"><br>
 <button type="submit">Register</button></form>
 <p><a href="/login">Back to sign in</a></p></div>{% endblock %}""",
     "forgot.html": """{% extends "base.html" %}{% block content %}
-<div class="card"><h2>Forgot password</h2>
+<div class="card auth"><h2>Forgot password</h2>
 <form method="post" action="/forgot">
 <input name="email" placeholder="email"><br>
 <button type="submit">Send reset token</button></form>
 <p><a href="/login">Back to sign in</a></p></div>{% endblock %}""",
     "reset.html": """{% extends "base.html" %}{% block content %}
-<div class="card"><h2>Reset password</h2>
+<div class="card auth"><h2>Reset password</h2>
 <form method="post" action="/reset">
 <input name="token" placeholder="reset token" value="{{ token }}"><br>
 <input name="password" type="password" placeholder="new password"><br>
@@ -275,6 +349,29 @@ monthly budget: {{ "%.2f"|format(budgets.get(p.id, 0) / 1000000) }} USD</p>
 
 _env = Environment(loader=DictLoader(_TEMPLATES), autoescape=select_autoescape(["html"]))
 
+#: template -> nav item highlighted as active
+_ACTIVE = {
+    "home.html": "/",
+    "warnings.html": "/warnings",
+    "scenarios.html": "/scenarios",
+    "runs.html": "/runs",
+    "run_detail.html": "/runs",
+    "playground.html": "/playground",
+    "datasets.html": "/datasets",
+    "dataset_detail.html": "/datasets",
+    "evals.html": "/evals",
+    "eval_detail.html": "/evals",
+    "prompts.html": "/prompts",
+    "prompt_detail.html": "/prompts",
+    "experiments.html": "/experiments",
+    "experiment_detail.html": "/experiments",
+    "agents.html": "/agents",
+    "projects.html": "/projects",
+    "health.html": "/health",
+}
+
 
 def render(name: str, ctx: Dict[str, Any]) -> str:
-    return _env.get_template(name).render(**ctx)
+    base = {"nav": _NAV, "active": _ACTIVE.get(name)}
+    base.update(ctx)
+    return _env.get_template(name).render(**base)
