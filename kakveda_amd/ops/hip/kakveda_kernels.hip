@@ -34,10 +34,18 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // epilogue is still partially exposed at 1 block/CU — the remaining
   // round-2 item (ROUND2.md).
   static const char* ksel = std::getenv("KAKVEDA_KNN_KERNEL");
+  // emission-epilogue re-entry guard: an overflowed emission run falls
+  // back to the list-epilogue path exactly once
+  static thread_local int emit_fallback = 0;
   const bool use8pbl = (ksel && std::string(ksel) == "8pbl") && N >= 4096;
   const bool use8pq = (ksel && std::string(ksel) == "8pq") && N >= 4096;
+  // 8pe: the 8-phase GEMM core with the threshold-emission epilogue +
+  // emit_merge_topk (no in-kernel lists). Needs the prepass floors, so
+  // only for corpora big enough to carry one (>= 64k columns).
+  const bool use8pe = (ksel && std::string(ksel) == "8pe") && N >= 65536 &&
+                      !emit_fallback;
   const bool use8p =
-      ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq) &&
+      ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
       N >= 4096;
   // Epilogue selection: default is EPI_MODE 11 (register-cached per-row
   // thresholds + ballot pre-check + inline single-insert fast path with
@@ -103,11 +111,16 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // KAKVEDA_KNN_PREPASS=0/1 forces it off/on.
   static const char* penv = std::getenv("KAKVEDA_KNN_PREPASS");
   const bool psmall = ntiles <= preg * PRE_TILES * 32;
-  const bool prepass = !use8p && k > 1 && preg >= 8 &&
-                       ntiles >= preg * PRE_TILES &&
-                       (penv ? penv[0] == '1' : psmall);
+  // the emission path REQUIRES the prepass: its published list minima are
+  // the exact per-row emission thresholds (see kernels_impl.h EPI_MODE 9)
+  const bool prepass = (use8pe || (!use8p && k > 1 &&
+                                   ntiles >= preg * PRE_TILES &&
+                                   (penv ? penv[0] == '1' : psmall))) &&
+                       preg >= 8;
   if (prepass) {
-    dim3 pgrid(preg, row_tiles);
+    // the prepass always runs the 128-row-tile kernel, also under the
+    // 256-row-tile 8p main launch
+    dim3 pgrid(preg, (B + BM - 1) / BM);
     if (epi == 11)
       hipLaunchKernelGGL((cosine_topk_partial_t<11>), pgrid, dim3(THREADS), 0, stream.stream(),
                          (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
@@ -140,6 +153,40 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)nullptr, (unsigned long long*)nullptr);
+  } else if (use8pe) {
+    // emission epilogue: candidates go to a per-row global buffer sized
+    // for ~8N/sample expected emissions with ~25x headroom
+    static const char* cenv = std::getenv("KAKVEDA_KNN_EMIT_CAP");
+    const long CAP = cenv ? std::atol(cenv) : 32768;
+    auto cand = torch::empty(
+        {(long)B, CAP},
+        torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
+    auto ccount = torch::zeros({B}, opts_i);
+    hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
+                       stream.stream(), (const bf16_t*)queries.data_ptr(),
+                       (const bf16_t*)corpus.data_ptr(),
+                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                       B, N, D, chunk_tiles, nchunks,
+                       (unsigned*)rowthr.data_ptr<int>(),
+                       (unsigned long long*)nullptr, (float*)nullptr,
+                       (unsigned long long*)cand.data_ptr<int64_t>(),
+                       (unsigned*)ccount.data_ptr<int>(), CAP);
+    hipLaunchKernelGGL(emit_merge_topk, dim3(B), dim3(256), 0, stream.stream(),
+                       (const unsigned long long*)cand.data_ptr<int64_t>(),
+                       (const unsigned*)ccount.data_ptr<int>(),
+                       out_score.data_ptr<float>(),
+                       (long*)out_idx.data_ptr<int64_t>(), B, CAP, (int)k);
+    // exactness guard (statistically never taken with prepass floors):
+    // an overflowed row means emission skipped stores — rerun the whole
+    // batch through the list-epilogue production kernel
+    const long mx = (long)ccount.max().item<int>();
+    if (mx > CAP) {
+      emit_fallback = 1;
+      auto r = cosine_topk(queries, corpus, k, valid_n);
+      emit_fallback = 0;
+      return r;
+    }
+    return {out_score, out_idx};
   } else if (use8pq) {
     hipLaunchKernelGGL((cosine_topk_partial8p_t<7>), grid, dim3(THREADS8), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),

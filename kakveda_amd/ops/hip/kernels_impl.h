@@ -701,13 +701,45 @@ constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 // the accumulators must stay live across the repair path — see
 // profiles/knn_kernel_history.md for why a queue-only version was
 // abandoned at the 256-VGPR/2-wave cap).
+// Rare-path candidate append for the emission epilogue (EPI_MODE 9).
+// __noinline__ on purpose: inlining its address math + store chain into
+// all 32 unrolled sweep groups pushes the 8p kernel past the 256-VGPR /
+// 2-wave cap into scratch (the historic register explosion); as a callee
+// it is only materialised once and only *called* on qualifying groups.
+__device__ __noinline__ void emit_candidates(
+    unsigned long long* __restrict__ cand, unsigned* __restrict__ ccount,
+    long ccap, int grow, float v0, float v1, float v2, float v3, int colb,
+    int N, float thr) {
+  const bool q0 = v0 >= thr && colb < N;
+  const bool q1 = v1 >= thr && colb + 16 < N;
+  const bool q2 = v2 >= thr && colb + 32 < N;
+  const bool q3 = v3 >= thr && colb + 48 < N;
+  const int myc = (int)q0 + (int)q1 + (int)q2 + (int)q3;
+  if (!myc) return;
+  unsigned pos = atomicAdd(&ccount[grow], (unsigned)myc);
+  unsigned long long* crow = cand + (size_t)grow * ccap;
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const bool qn = n == 0 ? q0 : n == 1 ? q1 : n == 2 ? q2 : q3;
+    const float vn = n == 0 ? v0 : n == 1 ? v1 : n == 2 ? v2 : v3;
+    if (qn) {
+      if (pos < (unsigned)ccap)
+        crow[pos] = ((unsigned long long)enc_f32(vn) << 32) |
+                    (unsigned)(0x7fffffff - (colb + n * 16));
+      ++pos;
+    }
+  }
+}
+
 template <int EPI_MODE>
 __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C,
     float* __restrict__ partial_score, int* __restrict__ partial_idx,
     int B, int N, int D, int chunk_tiles, int nchunks,
     unsigned* rowthr = nullptr, unsigned long long* stats = nullptr,
-    float* __restrict__ slab = nullptr) {
+    float* __restrict__ slab = nullptr,
+    unsigned long long* __restrict__ cand = nullptr,
+    unsigned* __restrict__ ccount = nullptr, long ccap = 0) {
   __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4 + 32 + BM8 * 4 + 32];
   char* const smem0 = smem;
   // buffer b in {0,1}: A half h at b*4*HALF8 + h*HALF8; B half h at +2*HALF8
@@ -876,10 +908,10 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
       if (rowthr != nullptr && tid < BM8 && row0 + tid < B)
         warm5 = dec_f32(rowthr[row0 + tid]);
     }
-    // EPI_MODE 6: per-lane threshold floors for this wave-half's two
+    // EPI_MODE 6/7/9: per-lane threshold floors for this wave-half's two
     // rows (rows beyond B get +inf so clamped-row garbage never flags)
     float thr0 = NEG_INF, thr1 = NEG_INF;
-    if constexpr (EPI_MODE == 6 || EPI_MODE == 7) {
+    if constexpr (EPI_MODE == 6 || EPI_MODE == 7 || EPI_MODE == 9) {
       const int r0g = row0 + wr * 128 + lane;
       thr0 = (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
       thr1 = (r0g + 64 < B) ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
@@ -980,6 +1012,37 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
+    } else if constexpr (EPI_MODE == 9) {
+      // ---- threshold-emission epilogue (round 2): no lists, no stash
+      // phases, no barriers — the 1017 TF GEMM core's full accumulator
+      // sweep is a register compare + rare global append. Exactness: the
+      // rowthr floors are prepass-published 8-deep list minima, i.e. the
+      // 8th-best of a score SUBSET, which lower-bounds the corpus
+      // 8th-best and hence the true k-th for any k <= 8 — so emitting
+      // every score >= floor provably captures the whole top-k. The
+      // companion emit_merge_topk kernel reduces the (expected ~8N/s per
+      // row) candidates; per-row counts above ccap flag a host fallback.
+      // pack (order-encoded score << 32) | (0x7fffffff - col): one u64
+      // max = higher score, then lower col (torch tie-break). The hot
+      // sweep is compare + ballot only; the append lives in the
+      // __noinline__ emit_candidates (see its comment).
+      const int colb = col0 + wc * 64 + cl;
+#pragma unroll
+      for (int m = 0; m < 8; ++m) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int rl = m * 16 + g * 4 + reg;  // 0..127 within this half
+          const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
+          const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                                   fmaxf(acc[m][2][reg], acc[m][3][reg]));
+          if (__ballot(gmax >= thr) == 0) continue;  // uniform: group dry
+          const int grow = row0 + wr * 128 + rl;
+          if (gmax >= thr && grow < B)
+            emit_candidates(cand, ccount, ccap, grow, acc[m][0][reg],
+                            acc[m][1][reg], acc[m][2][reg], acc[m][3][reg],
+                            colb, N, thr);
+        }
+      }
     } else {
       // ---- EPI_MODE 6/7 pre-check: one register ballot per (m,reg)
       // row group against the rowthr floors; waves with no qualifying
@@ -1240,17 +1303,103 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     }
   }
 
-  // write partials: [B][nchunks][KMAX]
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
-  if (tid < BM8) {
-    const int grow = row0 + tid;
-    if (grow < B) {
-      const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;
+  // write partials: [B][nchunks][KMAX] (emission mode has no lists — its
+  // results went straight to the candidate buffer)
+  if constexpr (EPI_MODE != 9) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (tid < BM8) {
+      const int grow = row0 + tid;
+      if (grow < B) {
+        const size_t base = ((size_t)grow * nchunks + chunk_id) * KMAX;
+#pragma unroll
+        for (int q = 0; q < KMAX; ++q) {
+          partial_score[base + q] = lsc[tid * KMAX + q];
+          partial_idx[base + q] = lix[tid * KMAX + q];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Companion merge for the emission epilogue (8p EPI_MODE 9): exact top-k
+// per row over the emitted (score, col) candidates. grid = B blocks x 256
+// threads; each thread keeps a sorted top-KMAX of its strided slice in
+// registers, then a log2(256)-step LDS tree merges them. Overflowed rows
+// (count > cap: emission skipped stores) write idx[0] = -2 so the host
+// falls back to the list-epilogue kernel — the exactness guard.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void emit_merge_topk(
+    const unsigned long long* __restrict__ cand,
+    const unsigned* __restrict__ ccount, float* __restrict__ out_s,
+    long* __restrict__ out_i, int B, long cap, int k) {
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const unsigned cnt = ccount[row];
+  if (cnt > (unsigned)cap) {
+    if (tid == 0) {
+      for (int q = 0; q < k; ++q) {
+        out_s[(size_t)row * k + q] = NEG_INF;
+        out_i[(size_t)row * k + q] = q == 0 ? -2 : -1;
+      }
+    }
+    return;
+  }
+  unsigned long long loc[KMAX];  // sorted desc; 0 = empty sentinel
+#pragma unroll
+  for (int q = 0; q < KMAX; ++q) loc[q] = 0ull;
+  const unsigned long long* crow = cand + (size_t)row * cap;
+  for (unsigned i = tid; i < cnt; i += 256) {
+    const unsigned long long v = crow[i];
+    if (v > loc[KMAX - 1]) {
+      loc[KMAX - 1] = v;  // insert at tail, bubble up (array stays sorted)
+#pragma unroll
+      for (int q = KMAX - 1; q > 0; --q)
+        if (loc[q] > loc[q - 1]) {
+          const unsigned long long t = loc[q];
+          loc[q] = loc[q - 1];
+          loc[q - 1] = t;
+        }
+    }
+  }
+  __shared__ unsigned long long all[256 * KMAX];
+#pragma unroll
+  for (int q = 0; q < KMAX; ++q) all[tid * KMAX + q] = loc[q];
+  __syncthreads();
+  for (int stride = 128; stride >= 1; stride >>= 1) {
+    if (tid < stride) {
+      // two-pointer merge of two sorted-desc KMAX lists -> top KMAX
+      // (LDS-indexed: a register array would go to scratch)
+      const unsigned long long* pa = all + (size_t)tid * KMAX;
+      const unsigned long long* pb = all + (size_t)(tid + stride) * KMAX;
+      unsigned long long o[KMAX];
+      int ia = 0, ib = 0;
 #pragma unroll
       for (int q = 0; q < KMAX; ++q) {
-        partial_score[base + q] = lsc[tid * KMAX + q];
-        partial_idx[base + q] = lix[tid * KMAX + q];
+        const unsigned long long va = pa[ia], vb = pb[ib];
+        if (va >= vb) {
+          o[q] = va;
+          ++ia;
+        } else {
+          o[q] = vb;
+          ++ib;
+        }
+      }
+#pragma unroll
+      for (int q = 0; q < KMAX; ++q) all[tid * KMAX + q] = o[q];
+    }
+    __syncthreads();
+  }
+  if (tid == 0) {
+    for (int q = 0; q < k; ++q) {
+      const unsigned long long v = all[q];
+      if (v == 0ull) {
+        out_s[(size_t)row * k + q] = NEG_INF;
+        out_i[(size_t)row * k + q] = -1;
+      } else {
+        out_s[(size_t)row * k + q] = dec_f32((unsigned)(v >> 32));
+        out_i[(size_t)row * k + q] = 0x7fffffffL - (long)(v & 0xffffffffu);
       }
     }
   }
@@ -1285,16 +1434,40 @@ __global__ __launch_bounds__(256) void kmeans_update_kernel(
   for (int i = tid; i < R * C * 64 + C; i += 256) part[i] = 0.f;
   __syncthreads();
 
-  // 4 points in flight per pass: thread (pl, d) = (tid>>6, tid&63)
+  // thread (pl, d) = (tid>>6, tid&63); 16 points batched per pass so the
+  // 2-byte strided loads stack up in flight (the 4-in-flight version was
+  // load-latency-bound at ~350 GB/s: each iteration's LDS add depended on
+  // its own just-issued load)
   const int pl = tid >> 6;
   const int d = tid & 63;
   float* mypart = part + (R == 4 ? pl * C * 64 : 0);
-  for (int p = p0 + pl; p < pend; p += 4) {
+  constexpr int PU = 16;
+  int p = p0 + pl;
+  for (; p + 4 * (PU - 1) < pend; p += 4 * PU) {
+    float v[PU];
+    int a[PU];
+#pragma unroll
+    for (int u = 0; u < PU; ++u) {
+      const int pp = p + 4 * u;
+      a[u] = assign[pp];
+      v[u] = (float)points[(size_t)pp * D + d0 + d];
+    }
+#pragma unroll
+    for (int u = 0; u < PU; ++u) {
+      if (a[u] >= 0 && a[u] < C) {
+        if (R == 4)
+          mypart[a[u] * 64 + d] += v[u];  // slot owned by this (pl, d) thread
+        else
+          atomicAdd(&part[a[u] * 64 + d], v[u]);
+      }
+    }
+  }
+  for (; p < pend; p += 4) {
     const int a = assign[p];
     const float v = (float)points[(size_t)p * D + d0 + d];
     if (a >= 0 && a < C) {
       if (R == 4)
-        mypart[a * 64 + d] += v;  // slot owned by this (pl, d) thread
+        mypart[a * 64 + d] += v;
       else
         atomicAdd(&part[a * 64 + d], v);
     }

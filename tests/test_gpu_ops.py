@@ -244,22 +244,25 @@ def test_k1_argmax_fast_path():
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("variant", ["eager", "fast", "rege", "8pbl"])
+@pytest.mark.parametrize("variant", ["eager", "fast", "rege", "8pbl", "8pe"])
 def test_kernel_variants_match_default(variant):
     """Every KAKVEDA_KNN_KERNEL variant must produce the same top-k
     scores as the default ballot kernel (env is read once per process,
-    so variants run in a subprocess)."""
+    so variants run in a subprocess). The emission variant (8pe) only
+    engages at N >= 64k (it needs the prepass floors), so it gets a
+    bigger corpus."""
     import os
     import subprocess
     import sys
 
+    n = 131072 if variant == "8pe" else 8192
     code = (
         "import torch\n"
         "from kakveda_amd import ops\n"
         "q = torch.randn(256, 768, generator=torch.Generator(device='cuda')"
         ".manual_seed(3), device='cuda')\n"
         "q = (q / q.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
-        "c = torch.randn(8192, 768, generator=torch.Generator(device='cuda')"
+        f"c = torch.randn({n}, 768, generator=torch.Generator(device='cuda')"
         ".manual_seed(4), device='cuda')\n"
         "c = (c / c.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
         "s, i = ops.cosine_topk(q, c, 5)\n"
@@ -371,3 +374,40 @@ def test_kmeans_assign_tie_breaks_to_first():
     _, idx = ops.kmeans_assign_scored(pts.contiguous(), cents.contiguous())
     torch.cuda.synchronize()
     assert (idx < 8).all(), idx.max()
+
+
+def test_emission_kernel_exact_vs_torch():
+    """The 8pe threshold-emission path (prepass floors + GEMM-core
+    emission + emit_merge_topk) is exact vs fp32 torch.topk, including a
+    valid_n prefix and k=8."""
+    import os
+    import subprocess
+    import sys
+
+    code = (
+        "import torch\n"
+        "from kakveda_amd import ops\n"
+        "g = torch.Generator(device='cuda').manual_seed(9)\n"
+        "q = torch.randn(300, 768, generator=g, device='cuda')\n"
+        "q = (q / q.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
+        "c = torch.randn(200001, 768, generator=g, device='cuda')\n"
+        "c = (c / c.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
+        "for k, vn in ((5, None), (8, None), (5, 150001), (1, None)):\n"
+        "    s, i = ops.cosine_topk(q, c, k, valid_n=vn)\n"
+        "    torch.cuda.synchronize()\n"
+        "    n = vn or c.shape[0]\n"
+        "    sims = q.float() @ c[:n].float().t()\n"
+        "    rs, ri = torch.topk(sims, k, dim=1)\n"
+        "    assert (i >= 0).all() and (i < n).all(), (k, vn)\n"
+        "    assert torch.allclose(s, rs, atol=2e-2, rtol=1e-2), (k, vn)\n"
+        "    gathered = sims.gather(1, i)\n"
+        "    assert torch.allclose(gathered, s, atol=1e-4), (k, vn)\n"
+        "    assert torch.allclose(gathered, rs, atol=1e-4), (k, vn)\n"
+        "print('OK8PE')\n"
+    )
+    env = dict(os.environ)
+    env["KAKVEDA_KNN_KERNEL"] = "8pe"
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "OK8PE" in r.stdout
