@@ -18,7 +18,7 @@ broadcast tensors.
 from __future__ import annotations
 
 import argparse
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.distributed as td
@@ -26,27 +26,43 @@ import torch.distributed as td
 from kakveda_amd.parallel.dist import init_from_env
 from kakveda_amd.parallel.sharded import ShardedStore
 
-CMD_APPEND = "append"
-CMD_SEARCH = "search"
-CMD_COUNT = "count"
-CMD_STOP = "stop"
+CMD_APPEND = 1
+CMD_SEARCH = 2
+CMD_STOP = 3
+
+#: fixed dtype codes for the command header (no pickling on the wire)
+_DTYPES = {0: torch.float32, 1: torch.bfloat16, 2: torch.float16}
+_DTYPE_IDS = {v: k for k, v in _DTYPES.items()}
 
 
 class DistGfkbCoordinator:
-    """Rank-0 handle: broadcasts a command, then performs the same local
-    collective call every worker performs."""
+    """Rank-0 handle: broadcasts a command header, then performs the same
+    local collective call every worker performs.
+
+    The command plane is ONE fixed-size int64[8] tensor broadcast
+    (cmd, n, d, k, dtype_id) followed by the payload broadcast — no
+    broadcast_object_list (which costs two broadcasts plus host-side
+    pickling per request). On GPUs the header
+    tensor rides RCCL like the payload.
+    """
 
     def __init__(self, dim: int = 768, capacity: int = 1 << 20):
         self.rank, self.world, self.device = init_from_env()
         self.dim = dim
         self.store = ShardedStore(dim, device=str(self.device), capacity=capacity)
+        self._hdr = torch.zeros(8, dtype=torch.int64, device=self.device)
 
     # -- plumbing ----------------------------------------------------------
 
-    def _bcast_cmd(self, cmd: str, meta: Tuple = ()) -> None:
+    def _bcast_cmd(self, cmd: int, n: int = 0, d: int = 0, k: int = 0,
+                   dtype: torch.dtype = torch.float32) -> None:
         if self.world > 1:
-            obj = [cmd, meta]
-            td.broadcast_object_list(obj, src=0)
+            self._hdr[0] = cmd
+            self._hdr[1] = n
+            self._hdr[2] = d
+            self._hdr[3] = k
+            self._hdr[4] = _DTYPE_IDS.get(dtype, 0)
+            td.broadcast(self._hdr, src=0)
 
     def _bcast_tensor(self, t: torch.Tensor) -> torch.Tensor:
         if self.world > 1:
@@ -57,13 +73,13 @@ class DistGfkbCoordinator:
 
     def append(self, rows: torch.Tensor) -> int:
         rows = rows.to(self.device)
-        self._bcast_cmd(CMD_APPEND, (rows.shape[0], rows.shape[1], str(rows.dtype)))
+        self._bcast_cmd(CMD_APPEND, rows.shape[0], rows.shape[1], dtype=rows.dtype)
         self._bcast_tensor(rows.contiguous())
         return self.store.append(rows)
 
     def search(self, queries: torch.Tensor, k: int):
         q = queries.to(self.device)
-        self._bcast_cmd(CMD_SEARCH, (q.shape[0], q.shape[1], int(k), str(q.dtype)))
+        self._bcast_cmd(CMD_SEARCH, q.shape[0], q.shape[1], int(k), dtype=q.dtype)
         self._bcast_tensor(q.contiguous())
         return self.store.search(q, k)
 
@@ -78,27 +94,23 @@ def worker_loop(coord_dim: int = 768, capacity: int = 1 << 20) -> None:
     """Ranks 1..N-1: execute broadcast commands until CMD_STOP."""
     rank, world, device = init_from_env()
     store = ShardedStore(coord_dim, device=str(device), capacity=capacity)
+    hdr = torch.zeros(8, dtype=torch.int64, device=device)
 
-    def recv_tensor(n: int, d: int, dtype_str: str) -> torch.Tensor:
-        dtype = getattr(torch, dtype_str.replace("torch.", ""))
+    def recv_tensor(n: int, d: int, dtype: torch.dtype) -> torch.Tensor:
         t = torch.empty(n, d, dtype=dtype, device=device)
         td.broadcast(t, src=0)
         return t
 
     while True:
-        obj = [None, None]
-        td.broadcast_object_list(obj, src=0)
-        cmd, meta = obj
+        td.broadcast(hdr, src=0)
+        cmd, n, d, k, dtid = (int(x) for x in hdr[:5].tolist())
         if cmd == CMD_STOP:
             break
+        dtype = _DTYPES.get(dtid, torch.float32)
         if cmd == CMD_APPEND:
-            n, d, dt = meta
-            store.append(recv_tensor(n, d, dt))
+            store.append(recv_tensor(n, d, dtype))
         elif cmd == CMD_SEARCH:
-            n, d, k, dt = meta
-            store.search(recv_tensor(n, d, dt), k)
-        elif cmd == CMD_COUNT:
-            pass
+            store.search(recv_tensor(n, d, dtype), k)
 
 
 def main(argv: Optional[list] = None) -> None:
