@@ -118,6 +118,12 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                                    (penv ? penv[0] == '1' : psmall))) &&
                        preg >= 8;
   if (prepass) {
+    if (use8pe) {
+      // the emission floor merge below reads every [row][chunk] partial
+      // slot; the prepass only writes chunks < preg, so blank the rest
+      pscore.fill_(-std::numeric_limits<float>::infinity());
+      pidx.fill_(-1);
+    }
     // the prepass always runs the 128-row-tile kernel, also under the
     // 256-row-tile 8p main launch
     dim3 pgrid(preg, (B + BM - 1) / BM);
@@ -162,6 +168,21 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
         {(long)B, CAP},
         torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
     auto ccount = torch::zeros({B}, opts_i);
+    // exact sample floor: merge the prepass partials into the true top-8
+    // of the whole sampled column set, publish its 8th as the emission
+    // threshold (see publish_emission_floor)
+    auto samp_s = torch::empty({(long)B, (long)KMAX}, opts_f);
+    auto samp_i = torch::empty(
+        {(long)B, (long)KMAX},
+        torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
+    hipLaunchKernelGGL(topk_merge_small, dim3((B + 255) / 256), dim3(256), 0,
+                       stream.stream(), pscore.data_ptr<float>(),
+                       pidx.data_ptr<int>(), samp_s.data_ptr<float>(),
+                       (long*)samp_i.data_ptr<int64_t>(), B, nchunks, KMAX);
+    hipLaunchKernelGGL(publish_emission_floor, dim3((B + 255) / 256),
+                       dim3(256), 0, stream.stream(),
+                       samp_s.data_ptr<float>(),
+                       (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
     hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
                        stream.stream(), (const bf16_t*)queries.data_ptr(),
                        (const bf16_t*)corpus.data_ptr(),
@@ -180,6 +201,10 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
     // an overflowed row means emission skipped stores — rerun the whole
     // batch through the list-epilogue production kernel
     const long mx = (long)ccount.max().item<int>();
+    static const char* denv = std::getenv("KAKVEDA_KNN_EMIT_DEBUG");
+    if (denv && denv[0] == '1')
+      printf("emit: rows=%d max=%ld mean=%.1f cap=%ld\n", B, mx,
+             ccount.to(torch::kFloat32).mean().item<float>(), CAP);
     if (mx > CAP) {
       emit_fallback = 1;
       auto r = cosine_topk(queries, corpus, k, valid_n);

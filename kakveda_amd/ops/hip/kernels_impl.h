@@ -1323,6 +1323,24 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 }
 
 // ---------------------------------------------------------------------------
+// Publish the merged sample's k-th best as the per-row emission floor.
+// After topk_merge over the prepass partials, out_s[row][kth] is the EXACT
+// kth-best of the whole sampled column set — a far tighter (still exact)
+// lower bound on the corpus kth-best than the per-(row, col-half) list
+// minima the prepass publishes on its own (those are 8th-of-512-columns;
+// at 10M rows they admitted ~40x more candidates and overflowed).
+// ---------------------------------------------------------------------------
+__global__ void publish_emission_floor(const float* __restrict__ merged_s,
+                                       unsigned* __restrict__ rowthr, int B,
+                                       int kth) {
+  const int r = blockIdx.x * blockDim.x + threadIdx.x;
+  if (r < B) {
+    const float v = merged_s[(size_t)r * KMAX + kth];
+    if (v > NEG_INF) atomicMax(&rowthr[r], enc_f32(v));
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Companion merge for the emission epilogue (8p EPI_MODE 9): exact top-k
 // per row over the emitted (score, col) candidates. grid = B blocks x 256
 // threads; each thread keeps a sorted top-KMAX of its strided slice in
