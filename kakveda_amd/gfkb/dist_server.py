@@ -11,8 +11,9 @@ Launch (8 GPUs):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \\
       --master-addr 127.0.0.1 -m kakveda_amd.gfkb.dist_server --port 8101
 
-Commands are tiny control tuples; payloads (embedding batches) travel as
-broadcast tensors.
+Commands are one fixed-size int64[8] header tensor broadcast; payloads
+(embedding/query batches) travel as tensor broadcasts — RCCL end to end
+on GPUs, no pickling.
 """
 
 from __future__ import annotations
