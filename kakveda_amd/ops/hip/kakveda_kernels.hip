@@ -42,8 +42,10 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // 8pe: the 8-phase GEMM core with the threshold-emission epilogue +
   // emit_merge_topk (no in-kernel lists). Needs the prepass floors, so
   // only for corpora big enough to carry one (>= 64k columns).
-  const bool use8pe = (ksel && std::string(ksel) == "8pe") && N >= 65536 &&
-                      !emit_fallback;
+  const bool use8pei = (ksel && std::string(ksel) == "8pei") && N >= 65536 &&
+                       !emit_fallback;  // inline-append emission variant
+  const bool use8pe = ((ksel && std::string(ksel) == "8pe") || use8pei) &&
+                      N >= 65536 && !emit_fallback;
   const bool use8p =
       ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
       N >= 4096;
@@ -183,15 +185,26 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        dim3(256), 0, stream.stream(),
                        samp_s.data_ptr<float>(),
                        (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
-    hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
-                       stream.stream(), (const bf16_t*)queries.data_ptr(),
-                       (const bf16_t*)corpus.data_ptr(),
-                       pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                       B, N, D, chunk_tiles, nchunks,
-                       (unsigned*)rowthr.data_ptr<int>(),
-                       (unsigned long long*)nullptr, (float*)nullptr,
-                       (unsigned long long*)cand.data_ptr<int64_t>(),
-                       (unsigned*)ccount.data_ptr<int>(), CAP);
+    if (use8pei)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<10>), grid, dim3(THREADS8),
+                         0, stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(),
+                         (unsigned long long*)nullptr, (float*)nullptr,
+                         (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), CAP);
+    else
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
+                         stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(),
+                         (unsigned long long*)nullptr, (float*)nullptr,
+                         (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), CAP);
     hipLaunchKernelGGL(emit_merge_topk, dim3(B), dim3(256), 0, stream.stream(),
                        (const unsigned long long*)cand.data_ptr<int64_t>(),
                        (const unsigned*)ccount.data_ptr<int>(),
