@@ -42,10 +42,8 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // 8pe: the 8-phase GEMM core with the threshold-emission epilogue +
   // emit_merge_topk (no in-kernel lists). Needs the prepass floors, so
   // only for corpora big enough to carry one (>= 64k columns).
-  const bool use8pei = (ksel && std::string(ksel) == "8pei") && N >= 65536 &&
-                       !emit_fallback;  // inline-append emission variant
-  const bool use8pe = ((ksel && std::string(ksel) == "8pe") || use8pei) &&
-                      N >= 65536 && !emit_fallback;
+  const bool use8pe = (ksel && std::string(ksel) == "8pe") && N >= 65536 &&
+                      !emit_fallback;
   const bool use8p =
       ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
       N >= 4096;
@@ -104,7 +102,13 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // writes land in slots the main launch overwrites).
   // grid.x must stay < nchunks so every pre-pass block's partial slot
   // [row][chunk_id] is in bounds (the main launch overwrites them all)
-  const int preg = std::min(64, nchunks) & ~7;
+  static const char* pgenv = std::getenv("KAKVEDA_KNN_PREG");
+  static const int preg_want = pgenv ? (int)std::atol(pgenv) : 0;
+  // emission floors tighten with sample size (E[emitted/row] ~ 8N/sample):
+  // default to a 4x bigger prepass for the emission path at large N
+  const int preg_base = use8pe ? 256 : 64;
+  const int preg =
+      std::min(preg_want > 0 ? preg_want : preg_base, nchunks) & ~7;
   constexpr int PRE_TILES = 8;
   // Pay the pre-pass only when the sample is a meaningful fraction of
   // the corpus (>= ~3%): below that its published floor is weaker than
@@ -185,18 +189,7 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        dim3(256), 0, stream.stream(),
                        samp_s.data_ptr<float>(),
                        (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
-    if (use8pei)
-      hipLaunchKernelGGL((cosine_topk_partial8p_t<10>), grid, dim3(THREADS8),
-                         0, stream.stream(), (const bf16_t*)queries.data_ptr(),
-                         (const bf16_t*)corpus.data_ptr(),
-                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                         B, N, D, chunk_tiles, nchunks,
-                         (unsigned*)rowthr.data_ptr<int>(),
-                         (unsigned long long*)nullptr, (float*)nullptr,
-                         (unsigned long long*)cand.data_ptr<int64_t>(),
-                         (unsigned*)ccount.data_ptr<int>(), CAP);
-    else
-      hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
+    hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
                          stream.stream(), (const bf16_t*)queries.data_ptr(),
                          (const bf16_t*)corpus.data_ptr(),
                          pscore.data_ptr<float>(), pidx.data_ptr<int>(),

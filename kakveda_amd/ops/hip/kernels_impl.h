@@ -1043,46 +1043,6 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
                             colb, N, thr);
         }
       }
-    } else if constexpr (EPI_MODE == 10) {
-      // emission with the append inlined and the sweep fully unrolled
-      // (a rolled m-loop dynamic-indexes acc -> 528 B/lane scratch):
-      // avoids the noinline callee's conservative entry waits, which
-      // drain the next tile's in-flight staging on every taken call
-      // (probe A/B vs mode 9)
-      const int colb = col0 + wc * 64 + cl;
-#pragma unroll
-      for (int m = 0; m < 8; ++m) {
-#pragma unroll
-        for (int reg = 0; reg < 4; ++reg) {
-          const int rl = m * 16 + g * 4 + reg;
-          const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
-          const float v0 = acc[m][0][reg], v1 = acc[m][1][reg];
-          const float v2 = acc[m][2][reg], v3 = acc[m][3][reg];
-          const float gmax = fmaxf(fmaxf(v0, v1), fmaxf(v2, v3));
-          if (__ballot(gmax >= thr) == 0) continue;
-          const int grow = row0 + wr * 128 + rl;
-          if (gmax >= thr && grow < B) {
-            const bool q0 = v0 >= thr && colb < N;
-            const bool q1 = v1 >= thr && colb + 16 < N;
-            const bool q2 = v2 >= thr && colb + 32 < N;
-            const bool q3 = v3 >= thr && colb + 48 < N;
-            const int myc = (int)q0 + (int)q1 + (int)q2 + (int)q3;
-            unsigned pos = atomicAdd(&ccount[grow], (unsigned)myc);
-            unsigned long long* crow = cand + (size_t)grow * ccap;
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              const bool qn = n == 0 ? q0 : n == 1 ? q1 : n == 2 ? q2 : q3;
-              const float vn = n == 0 ? v0 : n == 1 ? v1 : n == 2 ? v2 : v3;
-              if (qn) {
-                if (pos < (unsigned)ccap)
-                  crow[pos] = ((unsigned long long)enc_f32(vn) << 32) |
-                              (unsigned)(0x7fffffff - (colb + n * 16));
-                ++pos;
-              }
-            }
-          }
-        }
-      }
     } else {
       // ---- EPI_MODE 6/7 pre-check: one register ballot per (m,reg)
       // row group against the rowthr floors; waves with no qualifying
@@ -1343,8 +1303,8 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     }
   }
 
-  // write partials: [B][nchunks][KMAX] (emission mode has no lists — its
-  // results went straight to the candidate buffer)
+  // write partials: [B][nchunks][KMAX] (emission modes have no lists —
+  // their results went straight to the candidate buffer)
   if constexpr (EPI_MODE != 9) {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
