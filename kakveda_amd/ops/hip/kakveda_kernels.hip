@@ -123,39 +123,45 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                                    ntiles >= preg * PRE_TILES &&
                                    (penv ? penv[0] == '1' : psmall))) &&
                        preg >= 8;
+  // emission path: the prepass writes a COMPACT [B][preg][KMAX] partial
+  // buffer of its own (stride preg), so the floor merge scans only what
+  // the prepass produced instead of the whole [B][nchunks] layout
+  torch::Tensor ppre_s, ppre_i;
+  if (use8pe && prepass) {
+    ppre_s = torch::empty({(long)B * preg * KMAX}, opts_f);
+    ppre_i = torch::empty({(long)B * preg * KMAX}, opts_i);
+  }
   if (prepass) {
-    if (use8pe) {
-      // the emission floor merge below reads every [row][chunk] partial
-      // slot; the prepass only writes chunks < preg, so blank the rest
-      pscore.fill_(-std::numeric_limits<float>::infinity());
-      pidx.fill_(-1);
-    }
+    float* const pre_s =
+        use8pe ? ppre_s.data_ptr<float>() : pscore.data_ptr<float>();
+    int* const pre_i = use8pe ? ppre_i.data_ptr<int>() : pidx.data_ptr<int>();
+    const int pre_stride = use8pe ? preg : nchunks;
     // the prepass always runs the 128-row-tile kernel, also under the
     // 256-row-tile 8p main launch
     dim3 pgrid(preg, (B + BM - 1) / BM);
     if (epi == 11)
       hipLaunchKernelGGL((cosine_topk_partial_t<11>), pgrid, dim3(THREADS), 0, stream.stream(),
                          (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
-                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                         B, N, D, PRE_TILES, nchunks,
+                         pre_s, pre_i,
+                         B, N, D, PRE_TILES, pre_stride,
                          (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
     else if (epi == 7)
       hipLaunchKernelGGL((cosine_topk_partial_t<7>), pgrid, dim3(THREADS), 0, stream.stream(),
                          (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
-                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                         B, N, D, PRE_TILES, nchunks,
+                         pre_s, pre_i,
+                         B, N, D, PRE_TILES, pre_stride,
                          (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
     else if (epi == 0)
       hipLaunchKernelGGL((cosine_topk_partial_t<0>), pgrid, dim3(THREADS), 0, stream.stream(),
                          (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
-                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                         B, N, D, PRE_TILES, nchunks,
+                         pre_s, pre_i,
+                         B, N, D, PRE_TILES, pre_stride,
                          (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
     else
       hipLaunchKernelGGL((cosine_topk_partial_t<8>), pgrid, dim3(THREADS), 0, stream.stream(),
                          (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
-                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
-                         B, N, D, PRE_TILES, nchunks,
+                         pre_s, pre_i,
+                         B, N, D, PRE_TILES, pre_stride,
                          (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
   }
   if (k == 1 && !use8p) {
@@ -174,21 +180,25 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
         {(long)B, CAP},
         torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
     auto ccount = torch::zeros({B}, opts_i);
-    // exact sample floor: merge the prepass partials into the true top-8
-    // of the whole sampled column set, publish its 8th as the emission
-    // threshold (see publish_emission_floor)
+    // exact sample floor: merge the compact prepass partials into the
+    // true top-8 of the whole sampled column set, publish its 8th as the
+    // emission threshold (see publish_emission_floor). Block-per-row
+    // merge: the thread-per-row variant serial-scanned 2k entries per
+    // thread (0.9 ms/step at preg=256).
     auto samp_s = torch::empty({(long)B, (long)KMAX}, opts_f);
     auto samp_i = torch::empty(
         {(long)B, (long)KMAX},
         torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
-    hipLaunchKernelGGL(topk_merge_small, dim3((B + 255) / 256), dim3(256), 0,
-                       stream.stream(), pscore.data_ptr<float>(),
-                       pidx.data_ptr<int>(), samp_s.data_ptr<float>(),
-                       (long*)samp_i.data_ptr<int64_t>(), B, nchunks, KMAX);
-    hipLaunchKernelGGL(publish_emission_floor, dim3((B + 255) / 256),
-                       dim3(256), 0, stream.stream(),
-                       samp_s.data_ptr<float>(),
-                       (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
+    if (prepass) {
+      hipLaunchKernelGGL(topk_merge, dim3(B), dim3(THREADS), 0,
+                         stream.stream(), ppre_s.data_ptr<float>(),
+                         ppre_i.data_ptr<int>(), samp_s.data_ptr<float>(),
+                         (long*)samp_i.data_ptr<int64_t>(), preg, KMAX);
+      hipLaunchKernelGGL(publish_emission_floor, dim3((B + 255) / 256),
+                         dim3(256), 0, stream.stream(),
+                         samp_s.data_ptr<float>(),
+                         (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
+    }
     hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
                          stream.stream(), (const bf16_t*)queries.data_ptr(),
                          (const bf16_t*)corpus.data_ptr(),

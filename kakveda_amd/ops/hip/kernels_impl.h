@@ -1023,10 +1023,17 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
       // companion emit_merge_topk kernel reduces the (expected ~8N/s per
       // row) candidates; per-row counts above ccap flag a host fallback.
       // pack (order-encoded score << 32) | (0x7fffffff - col): one u64
-      // max = higher score, then lower col (torch tie-break). The hot
-      // sweep is compare + ballot only; the append lives in the
-      // __noinline__ emit_candidates (see its comment).
+      // max = higher score, then lower col (torch tie-break).
+      // Two-phase sweep: the HOT pass is pure compare/shfl/ballot with
+      // NO call sites, so the compiler keeps the 128 accumulator VGPRs
+      // unspilled (a single-pass sweep with per-group calls pre-spilled
+      // 32 regs per tile per wave -> ~27 GB of scratch traffic at the
+      // 10M bench, the measured 917-vs-1017 TF gap). The COLD pass
+      // (wave-uniform qmask != 0, ~12% of tiles per wave) re-derives the
+      // thresholds and makes the rare __noinline__ emit_candidates
+      // calls; its spills live in that cold block only.
       const int colb = col0 + wc * 64 + cl;
+      unsigned qm32 = 0;
 #pragma unroll
       for (int m = 0; m < 8; ++m) {
 #pragma unroll
@@ -1035,12 +1042,26 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
           const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
           const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
                                    fmaxf(acc[m][2][reg], acc[m][3][reg]));
-          if (__ballot(gmax >= thr) == 0) continue;  // uniform: group dry
-          const int grow = row0 + wr * 128 + rl;
-          if (gmax >= thr && grow < B)
-            emit_candidates(cand, ccount, ccap, grow, acc[m][0][reg],
-                            acc[m][1][reg], acc[m][2][reg], acc[m][3][reg],
-                            colb, N, thr);
+          if (__ballot(gmax >= thr))  // uniform
+            qm32 |= 1u << (m * 4 + reg);
+        }
+      }
+      if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
+#pragma unroll
+        for (int m = 0; m < 8; ++m) {
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) {
+            if (!(qm32 & (1u << (m * 4 + reg)))) continue;
+            const int rl = m * 16 + g * 4 + reg;
+            const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
+            const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                                     fmaxf(acc[m][2][reg], acc[m][3][reg]));
+            const int grow = row0 + wr * 128 + rl;
+            if (gmax >= thr && grow < B)
+              emit_candidates(cand, ccount, ccap, grow, acc[m][0][reg],
+                              acc[m][1][reg], acc[m][2][reg], acc[m][3][reg],
+                              colb, N, thr);
+          }
         }
       }
     } else {
