@@ -403,6 +403,12 @@ def test_emission_kernel_exact_vs_torch():
         "    gathered = sims.gather(1, i)\n"
         "    assert torch.allclose(gathered, s, atol=1e-4), (k, vn)\n"
         "    assert torch.allclose(gathered, rs, atol=1e-4), (k, vn)\n"
+        "# B=1 single-query serving shape through the emission path\n"
+        "s1, i1 = ops.cosine_topk(q[:1].contiguous(), c, 5)\n"
+        "torch.cuda.synchronize()\n"
+        "sims1 = q[:1].float() @ c.float().t()\n"
+        "rs1, _ = torch.topk(sims1, 5, dim=1)\n"
+        "assert torch.allclose(sims1.gather(1, i1), rs1, atol=1e-4)\n"
         "print('OK8PE')\n"
     )
     env = dict(os.environ)
