@@ -96,11 +96,8 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   dim3 grid(nchunks, row_tiles);
   hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0,
                      stream.stream(), (unsigned*)rowthr.data_ptr<int>(), B);
-  torch::Tensor slab;
-  if (use8p) {
-    slab = torch::empty({(long)nchunks * row_tiles * BM8 * BN8},
-                        torch::TensorOptions().dtype(torch::kFloat32).device(queries.device()));
-  }
+  // (the EPI_MODE 5 score slab is probe-only; no host path passes it, so
+  // nothing allocates the ~650 MB/call buffer anymore)
   // Threshold pre-pass: one cheap launch over the first PRE_TILES*PREG
   // column tiles fills each row's lists from a ~4k-column sample and
   // publishes their minima into rowthr, so the main launch's blocks all
@@ -242,21 +239,21 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
-    (void)slab;
+    // (slab: probe-only EPI_MODE 5 path)
   } else if (use8pbl) {
     hipLaunchKernelGGL((cosine_topk_partial8p_t<6>), grid, dim3(THREADS8), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
-    (void)slab;
+    // (slab: probe-only EPI_MODE 5 path)
   } else if (use8p) {
     hipLaunchKernelGGL((cosine_topk_partial8p_t<0>), grid, dim3(THREADS8), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
                        pscore.data_ptr<float>(), pidx.data_ptr<int>(),
                        B, N, D, chunk_tiles, nchunks,
                        (unsigned*)rowthr.data_ptr<int>(), (unsigned long long*)nullptr);
-    (void)slab;
+    // (slab: probe-only EPI_MODE 5 path)
   } else if (epi == 7) {
     hipLaunchKernelGGL((cosine_topk_partial_t<7>), grid, dim3(THREADS), 0, stream.stream(),
                        (const bf16_t*)queries.data_ptr(), (const bf16_t*)corpus.data_ptr(),
