@@ -412,6 +412,72 @@ std::tuple<torch::Tensor, torch::Tensor> kmeans_update(
   return {sums, counts};
 }
 
+double probe8p(torch::Tensor queries, torch::Tensor corpus, int64_t mode,
+               int64_t iters) {
+  // Timing probe for 8p epilogue isolation (profiles/knn_kernel_history):
+  // mode 1 = no epilogue (GEMM core), 12 = hot sweep only (worst case:
+  // null thresholds make every ballot fire), 9 = full emission with
+  // -inf floors (bootstrap worst case). Returns ms per launch.
+  check_bf16_2d(queries, "queries");
+  check_bf16_2d(corpus, "corpus");
+  const int B = queries.size(0);
+  const int D = queries.size(1);
+  const int N = corpus.size(0);
+  const int row_tiles = (B + BM8 - 1) / BM8;
+  const int ntiles = (N + BN8 - 1) / BN8;
+  long want = ((long)ntiles * row_tiles + 511) / 512;
+  const int chunk_tiles = (int)std::max(4L, std::min(want, 128L));
+  const int nchunks = ((ntiles + chunk_tiles - 1) / chunk_tiles + 7) & ~7;
+  auto opts_f = torch::TensorOptions().dtype(torch::kFloat32).device(queries.device());
+  auto opts_i = torch::TensorOptions().dtype(torch::kInt32).device(queries.device());
+  auto pscore = torch::empty({(long)B * nchunks * KMAX}, opts_f);
+  auto pidx = torch::empty({(long)B * nchunks * KMAX}, opts_i);
+  auto cand = torch::empty({(long)B, 65536},
+                           torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
+  auto ccount = torch::zeros({B}, opts_i);
+  auto stream = c10::hip::getCurrentHIPStream();
+  dim3 grid(nchunks, row_tiles);
+  auto launch = [&](int md) {
+    if (md == 1)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<1>), grid, dim3(THREADS8), 0,
+                         stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(), pscore.data_ptr<float>(),
+                         pidx.data_ptr<int>(), B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)nullptr, (unsigned long long*)nullptr,
+                         (float*)nullptr, (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), 65536);
+    else if (md == 12)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<12>), grid, dim3(THREADS8), 0,
+                         stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(), pscore.data_ptr<float>(),
+                         pidx.data_ptr<int>(), B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)nullptr, (unsigned long long*)nullptr,
+                         (float*)nullptr, (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), 65536);
+    else
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
+                         stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(), pscore.data_ptr<float>(),
+                         pidx.data_ptr<int>(), B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)nullptr, (unsigned long long*)nullptr,
+                         (float*)nullptr, (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), 65536);
+  };
+  launch((int)mode);  // warmup
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  hipEventRecord(e0, stream.stream());
+  for (long i = 0; i < iters; ++i) launch((int)mode);
+  hipEventRecord(e1, stream.stream());
+  hipEventSynchronize(e1);
+  float ms = 0.f;
+  hipEventElapsedTime(&ms, e0, e1);
+  hipEventDestroy(e0);
+  hipEventDestroy(e1);
+  return (double)ms / (double)iters;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cosine_topk", &cosine_topk, "fused cosine top-k (MFMA + LDS top-k)");
   m.def("l2normalize_", &l2normalize_, "in-place row L2 normalisation");
@@ -419,5 +485,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_update", &kmeans_update, "segmented centroid sum + counts");
   m.def("kmeans_assign", &kmeans_assign,
         "argmax-cosine assignment, LDS-resident centroids (C<=64)");
+  m.def("probe8p", &probe8p, "8p epilogue-isolation timing probe");
   m.attr("KMAX") = KMAX;
 }

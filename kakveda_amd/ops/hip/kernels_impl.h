@@ -911,7 +911,8 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     // EPI_MODE 6/7/9: per-lane threshold floors for this wave-half's two
     // rows (rows beyond B get +inf so clamped-row garbage never flags)
     float thr0 = NEG_INF, thr1 = NEG_INF;
-    if constexpr (EPI_MODE == 6 || EPI_MODE == 7 || EPI_MODE == 9) {
+    if constexpr (EPI_MODE == 6 || EPI_MODE == 7 || EPI_MODE == 9 ||
+                  EPI_MODE == 12) {
       const int r0g = row0 + wr * 128 + lane;
       thr0 = (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
       thr1 = (r0g + 64 < B) ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
@@ -1012,7 +1013,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
-    } else if constexpr (EPI_MODE == 9) {
+    } else if constexpr (EPI_MODE == 9 || EPI_MODE == 12) {
       // ---- threshold-emission epilogue (round 2): no lists, no stash
       // phases, no barriers — the 1017 TF GEMM core's full accumulator
       // sweep is a register compare + rare global append. Exactness: the
@@ -1046,7 +1047,11 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
             qm32 |= 1u << (m * 4 + reg);
         }
       }
-      if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
+      if constexpr (EPI_MODE == 12) {
+        // probe-only: hot sweep computed, cold path compiled out — lets
+        // the host isolate the sweep's cost from the emission cost
+        asm volatile("" ::"s"(qm32));
+      } else if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
 #pragma unroll
         for (int m = 0; m < 8; ++m) {
 #pragma unroll
@@ -1326,7 +1331,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 
   // write partials: [B][nchunks][KMAX] (emission modes have no lists —
   // their results went straight to the candidate buffer)
-  if constexpr (EPI_MODE != 9) {
+  if constexpr (EPI_MODE != 9 && EPI_MODE != 12) {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     if (tid < BM8) {
