@@ -701,32 +701,45 @@ constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 // the accumulators must stay live across the repair path — see
 // profiles/knn_kernel_history.md for why a queue-only version was
 // abandoned at the 256-VGPR/2-wave cap).
-// Rare-path candidate append for the emission epilogue (EPI_MODE 9).
-// __noinline__ on purpose: inlining its address math + store chain into
-// all 32 unrolled sweep groups pushes the 8p kernel past the 256-VGPR /
-// 2-wave cap into scratch (the historic register explosion); as a callee
-// it is only materialised once and only *called* on qualifying groups.
-__device__ __noinline__ void emit_candidates(
+// Rare-path drain for the emission epilogue (EPI_MODE 9): the cold path
+// stashes qualifying groups' accumulator quads into the wave's private
+// slice of the dead A-image (plain LDS stores, no calls while the 128
+// accumulator VGPRs are live), then makes ONE call here per wave per
+// cold tile. By call time the accumulators are dead, so the call costs
+// no spills — earlier structures (per-group callee: 32-reg/tile
+// pre-spill; inlined appends: 0.5-1 KiB/lane scratch) all lost to the
+// acc-liveness-across-append problem. __noinline__ keeps the body out
+// of the sweep's register allocation entirely.
+__device__ __noinline__ void emit_stashed(
     unsigned long long* __restrict__ cand, unsigned* __restrict__ ccount,
-    long ccap, int grow, float v0, float v1, float v2, float v3, int colb,
-    int N, float thr) {
-  const bool q0 = v0 >= thr && colb < N;
-  const bool q1 = v1 >= thr && colb + 16 < N;
-  const bool q2 = v2 >= thr && colb + 32 < N;
-  const bool q3 = v3 >= thr && colb + 48 < N;
-  const int myc = (int)q0 + (int)q1 + (int)q2 + (int)q3;
-  if (!myc) return;
-  unsigned pos = atomicAdd(&ccount[grow], (unsigned)myc);
-  unsigned long long* crow = cand + (size_t)grow * ccap;
+    long ccap, char* stash, int ng, int rowbase, int colb, int N, int B,
+    float thr0, float thr1) {
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;
+  for (int i = 0; i < ng; ++i) {
+    const int gid = (int)((volatile unsigned*)stash)[i];
+    const int m = gid >> 2, reg = gid & 3;
+    const int rl = m * 16 + g * 4 + reg;
+    const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
+    const f32x4 v = *(const f32x4*)(stash + 64 + i * 1024 + lane * 16);
+    const int grow = rowbase + rl;
+    const bool q0 = v[0] >= thr && colb < N;
+    const bool q1 = v[1] >= thr && colb + 16 < N;
+    const bool q2 = v[2] >= thr && colb + 32 < N;
+    const bool q3 = v[3] >= thr && colb + 48 < N;
+    const int myc = (int)q0 + (int)q1 + (int)q2 + (int)q3;
+    if (!myc || grow >= B) continue;
+    unsigned pos = atomicAdd(&ccount[grow], (unsigned)myc);
+    unsigned long long* crow = cand + (size_t)grow * ccap;
 #pragma unroll
-  for (int n = 0; n < 4; ++n) {
-    const bool qn = n == 0 ? q0 : n == 1 ? q1 : n == 2 ? q2 : q3;
-    const float vn = n == 0 ? v0 : n == 1 ? v1 : n == 2 ? v2 : v3;
-    if (qn) {
-      if (pos < (unsigned)ccap)
-        crow[pos] = ((unsigned long long)enc_f32(vn) << 32) |
-                    (unsigned)(0x7fffffff - (colb + n * 16));
-      ++pos;
+    for (int n = 0; n < 4; ++n) {
+      const bool qn = n == 0 ? q0 : n == 1 ? q1 : n == 2 ? q2 : q3;
+      if (qn) {
+        if (pos < (unsigned)ccap)
+          crow[pos] = ((unsigned long long)enc_f32(v[n]) << 32) |
+                      (unsigned)(0x7fffffff - (colb + n * 16));
+        ++pos;
+      }
     }
   }
 }
@@ -1051,23 +1064,46 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         // probe-only: hot sweep computed, cold path compiled out — lets
         // the host isolate the sweep's cost from the emission cost
         asm volatile("" ::"s"(qm32));
-      } else if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
+      } else {
+        if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
+          // stash qualifying groups into the wave's 4 KiB slice of the
+          // dead last-window A-image: plain LDS stores only (see
+          // emit_stashed's comment for why no call may happen while the
+          // accumulators are live)
+          const int t_laste = j * nkt + nkt - 1;
+          char* stash = ahalf(t_laste & 1, 0) + wid * 4096;
+          int ng = 0;
 #pragma unroll
-        for (int m = 0; m < 8; ++m) {
+          for (int m = 0; m < 8; ++m) {
 #pragma unroll
-          for (int reg = 0; reg < 4; ++reg) {
-            if (!(qm32 & (1u << (m * 4 + reg)))) continue;
-            const int rl = m * 16 + g * 4 + reg;
-            const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
-            const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
-                                     fmaxf(acc[m][2][reg], acc[m][3][reg]));
-            const int grow = row0 + wr * 128 + rl;
-            if (gmax >= thr && grow < B)
-              emit_candidates(cand, ccount, ccap, grow, acc[m][0][reg],
-                              acc[m][1][reg], acc[m][2][reg], acc[m][3][reg],
-                              colb, N, thr);
+            for (int reg = 0; reg < 4; ++reg) {
+              if (!(qm32 & (1u << (m * 4 + reg)))) continue;
+              if (ng < 3) {
+                *(f32x4*)(stash + 64 + ng * 1024 + (size_t)lane * 16) =
+                    f32x4{acc[m][0][reg], acc[m][1][reg], acc[m][2][reg],
+                          acc[m][3][reg]};
+                if (lane == 0)
+                  ((volatile unsigned*)stash)[ng] = (unsigned)(m * 4 + reg);
+              }
+              ++ng;
+            }
+          }
+          if (__builtin_expect(ng > 3, 0)) {
+            // >3 qualifying groups in one wave-tile (possible only with
+            // degenerate floors): poison a row count -> host fallback
+            if (lane == 0)
+              atomicAdd(&ccount[row0 < B ? row0 : B - 1],
+                        (unsigned)(ccap + 1));
+          } else {
+            emit_stashed(cand, ccount, ccap, stash, ng, row0 + wr * 128,
+                         colb, N, B, thr0, thr1);
           }
         }
+        // every wave barriers so the next tile's A staging cannot land in
+        // a still-draining wave's stash slice (the slice spans another
+        // wave's staging quadrant)
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
       }
     } else {
       // ---- EPI_MODE 6/7 pre-check: one register ballot per (m,reg)
