@@ -42,16 +42,14 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // 8pe: the 8-phase GEMM core with the threshold-emission epilogue +
   // emit_merge_topk (no in-kernel lists). Needs the prepass floors, so
   // only for corpora big enough to carry one (>= 64k columns).
-  // Default ON for 64k <= N <= 4M (k > 1): measured +17-21% at 1M and
-  // +14% at 2M over the ballot kernel, crossover ~4-5M, -3% at 6M+
-  // (the 32-group sweep costs ~10% of the kernel; the 8p GEMM-core
-  // advantage only outruns it below the crossover — see
-  // profiles/knn_kernel_history.md round 2). KAKVEDA_KNN_KERNEL=8pe
-  // forces it for any N >= 64k; any other explicit selection disables.
+  // DEFAULT for every N >= 64k with k > 1 since the stash-drain cold
+  // path went spill-free: same-box A/B measured +26% at 1M, +22% at 2M,
+  // parity at 6M and +4.5% at 10M over the ballot kernel
+  // (profiles/knn_kernel_history.md round 2). Any explicit
+  // KAKVEDA_KNN_KERNEL selection other than 8pe disables it.
   const bool use8pe =
-      ((ksel ? std::string(ksel) == "8pe"
-             : (k > 1 && N <= 4000000))) &&
-      N >= 65536 && !emit_fallback;
+      (ksel ? std::string(ksel) == "8pe" : k > 1) && N >= 65536 &&
+      !emit_fallback;
   const bool use8p =
       ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
       N >= 4096;
