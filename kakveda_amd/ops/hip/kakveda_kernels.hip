@@ -222,8 +222,9 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
     const long mx = (long)ccount.max().item<int>();
     static const char* denv = std::getenv("KAKVEDA_KNN_EMIT_DEBUG");
     if (denv && denv[0] == '1')
-      printf("emit: rows=%d max=%ld mean=%.1f cap=%ld\n", B, mx,
-             ccount.to(torch::kFloat32).mean().item<float>(), CAP);
+      printf("emit: rows=%d max=%ld mean=%.1f cap=%ld%s\n", B, mx,
+             ccount.to(torch::kFloat32).mean().item<float>(), CAP,
+             mx > CAP ? "  FALLBACK" : "");
     if (mx > CAP) {
       emit_fallback = 1;
       auto r = cosine_topk(queries, corpus, k, valid_n);
