@@ -226,6 +226,14 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
              ccount.to(torch::kFloat32).mean().item<float>(), CAP,
              mx > CAP ? "  FALLBACK" : "");
     if (mx > CAP) {
+      static bool warned = false;
+      if (!warned) {
+        warned = true;
+        fprintf(stderr,
+                "[kakveda] emission candidate overflow (max=%ld > cap=%ld) — "
+                "falling back to the list-epilogue kernel for this batch\n",
+                mx, CAP);
+      }
       emit_fallback = 1;
       auto r = cosine_topk(queries, corpus, k, valid_n);
       emit_fallback = 0;

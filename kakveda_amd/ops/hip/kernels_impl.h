@@ -1099,10 +1099,17 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
                          colb, N, B, thr0, thr1);
           }
         }
-        // every wave barriers so the next tile's A staging cannot land in
+        // Every wave barriers so the next tile's A staging cannot land in
         // a still-draining wave's stash slice (the slice spans another
-        // wave's staging quadrant)
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        // wave's staging quadrant). The vmcnt(0) is LOAD-BEARING: the
+        // drain's global atomics/stores are vmcnt-tracked on gfx9-family,
+        // and any left outstanding here would be counted by the next
+        // window's `vmcnt(4)` staging certification — which would then
+        // "prove" A/B tiles that have NOT landed and read uncertified
+        // LDS (garbage scores -> emission storms -> silent fallback;
+        // manifested box-dependently, the same counted-wait audit class
+        // as the round-1 chunk-tail race).
+        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();
       }
     } else {
@@ -1352,6 +1359,10 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();  // lists settled; stash reusable
       }
+      // drain the drains' rowthr publish atomics: they are vmcnt-tracked
+      // and must not leak into the next tile's counted staging waits
+      // (same audit class as the emission epilogue's vmcnt(0))
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
   }
 
