@@ -419,6 +419,47 @@ std::tuple<torch::Tensor, torch::Tensor> kmeans_update(
   return {sums, counts};
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> floor_probe(
+    torch::Tensor queries, torch::Tensor corpus, int64_t pregq) {
+  // Debug: run ONLY the emission floor pipeline (prepass -> compact
+  // partials -> merge -> publish) and return (samp_s [B,KMAX], samp_i,
+  // rowthr u32 [B]) for offline comparison against a torch reference.
+  check_bf16_2d(queries, "queries");
+  check_bf16_2d(corpus, "corpus");
+  const int B = queries.size(0);
+  const int D = queries.size(1);
+  const int N = corpus.size(0);
+  const int preg = ((int)pregq) & ~7;
+  auto opts_f = torch::TensorOptions().dtype(torch::kFloat32).device(queries.device());
+  auto opts_i = torch::TensorOptions().dtype(torch::kInt32).device(queries.device());
+  auto rowthr = torch::empty({B}, opts_i);
+  auto ppre_s = torch::empty({(long)B * preg * KMAX}, opts_f);
+  auto ppre_i = torch::empty({(long)B * preg * KMAX}, opts_i);
+  auto samp_s = torch::empty({(long)B, (long)KMAX}, opts_f);
+  auto samp_i = torch::empty(
+      {(long)B, (long)KMAX},
+      torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0,
+                     stream.stream(), (unsigned*)rowthr.data_ptr<int>(), B);
+  constexpr int PRE_TILES = 8;
+  dim3 pgrid(preg, (B + BM - 1) / BM);
+  hipLaunchKernelGGL((cosine_topk_partial_t<11>), pgrid, dim3(THREADS), 0,
+                     stream.stream(), (const bf16_t*)queries.data_ptr(),
+                     (const bf16_t*)corpus.data_ptr(), ppre_s.data_ptr<float>(),
+                     ppre_i.data_ptr<int>(), B, N, D, PRE_TILES, preg,
+                     (unsigned*)rowthr.data_ptr<int>(),
+                     (unsigned long long*)nullptr);
+  hipLaunchKernelGGL(topk_merge, dim3(B), dim3(THREADS), 0, stream.stream(),
+                     ppre_s.data_ptr<float>(), ppre_i.data_ptr<int>(),
+                     samp_s.data_ptr<float>(),
+                     (long*)samp_i.data_ptr<int64_t>(), preg, KMAX);
+  hipLaunchKernelGGL(publish_emission_floor, dim3((B + 255) / 256), dim3(256),
+                     0, stream.stream(), samp_s.data_ptr<float>(),
+                     (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
+  return {samp_s, samp_i, rowthr};
+}
+
 double probe8p(torch::Tensor queries, torch::Tensor corpus, int64_t mode,
                int64_t iters) {
   // Timing probe for 8p epilogue isolation (profiles/knn_kernel_history):
@@ -493,5 +534,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_assign", &kmeans_assign,
         "argmax-cosine assignment, LDS-resident centroids (C<=64)");
   m.def("probe8p", &probe8p, "8p epilogue-isolation timing probe");
+  m.def("floor_probe", &floor_probe, "emission floor pipeline debug probe");
   m.attr("KMAX") = KMAX;
 }
