@@ -710,6 +710,33 @@ constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
 // pre-spill; inlined appends: 0.5-1 KiB/lane scratch) all lost to the
 // acc-liveness-across-append problem. __noinline__ keeps the body out
 // of the sweep's register allocation entirely.
+// v3-era per-group append callee, kept for A/B isolation (EPI_MODE 14):
+// calls it once per qualifying group with the accumulators still live.
+__device__ __noinline__ void emit_candidates(
+    unsigned long long* __restrict__ cand, unsigned* __restrict__ ccount,
+    long ccap, int grow, float v0, float v1, float v2, float v3, int colb,
+    int N, float thr) {
+  const bool q0 = v0 >= thr && colb < N;
+  const bool q1 = v1 >= thr && colb + 16 < N;
+  const bool q2 = v2 >= thr && colb + 32 < N;
+  const bool q3 = v3 >= thr && colb + 48 < N;
+  const int myc = (int)q0 + (int)q1 + (int)q2 + (int)q3;
+  if (!myc) return;
+  unsigned pos = atomicAdd(&ccount[grow], (unsigned)myc);
+  unsigned long long* crow = cand + (size_t)grow * ccap;
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const bool qn = n == 0 ? q0 : n == 1 ? q1 : n == 2 ? q2 : q3;
+    const float vn = n == 0 ? v0 : n == 1 ? v1 : n == 2 ? v2 : v3;
+    if (qn) {
+      if (pos < (unsigned)ccap)
+        crow[pos] = ((unsigned long long)enc_f32(vn) << 32) |
+                    (unsigned)(0x7fffffff - (colb + n * 16));
+      ++pos;
+    }
+  }
+}
+
 __device__ __noinline__ void emit_stashed(
     unsigned long long* __restrict__ cand, unsigned* __restrict__ ccount,
     long ccap, char* stash, int ng, int rowbase, int colb, int N, int B,
@@ -925,7 +952,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     // rows (rows beyond B get +inf so clamped-row garbage never flags)
     float thr0 = NEG_INF, thr1 = NEG_INF;
     if constexpr (EPI_MODE == 6 || EPI_MODE == 7 || EPI_MODE == 9 ||
-                  EPI_MODE == 12) {
+                  EPI_MODE == 12 || EPI_MODE == 14) {
       const int r0g = row0 + wr * 128 + lane;
       thr0 = (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
       thr1 = (r0g + 64 < B) ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
@@ -1026,7 +1053,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
-    } else if constexpr (EPI_MODE == 9 || EPI_MODE == 12) {
+    } else if constexpr (EPI_MODE == 9 || EPI_MODE == 12 || EPI_MODE == 14) {
       // ---- threshold-emission epilogue (round 2): no lists, no stash
       // phases, no barriers — the 1017 TF GEMM core's full accumulator
       // sweep is a register compare + rare global append. Exactness: the
@@ -1064,6 +1091,27 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         // probe-only: hot sweep computed, cold path compiled out — lets
         // the host isolate the sweep's cost from the emission cost
         asm volatile("" ::"s"(qm32));
+      } else if constexpr (EPI_MODE == 14) {
+        // isolation A/B: the v3-era cold path (per-group callee with the
+        // accumulators live; pays the pre-spill, no stash, no barrier)
+        if (__builtin_expect(qm32 != 0, 0)) {
+#pragma unroll
+          for (int m = 0; m < 8; ++m) {
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+              if (!(qm32 & (1u << (m * 4 + reg)))) continue;
+              const int rl = m * 16 + g * 4 + reg;
+              const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
+              const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                                       fmaxf(acc[m][2][reg], acc[m][3][reg]));
+              const int grow = row0 + wr * 128 + rl;
+              if (gmax >= thr && grow < B)
+                emit_candidates(cand, ccount, ccap, grow, acc[m][0][reg],
+                                acc[m][1][reg], acc[m][2][reg],
+                                acc[m][3][reg], colb, N, thr);
+            }
+          }
+        }
       } else {
         if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
           // stash qualifying groups into the wave's 4 KiB slice of the
@@ -1378,7 +1426,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 
   // write partials: [B][nchunks][KMAX] (emission modes have no lists —
   // their results went straight to the candidate buffer)
-  if constexpr (EPI_MODE != 9 && EPI_MODE != 12) {
+  if constexpr (EPI_MODE != 9 && EPI_MODE != 12 && EPI_MODE != 14) {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     if (tid < BM8) {
