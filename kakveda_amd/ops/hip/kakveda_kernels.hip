@@ -47,9 +47,11 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // parity at 6M and +4.5% at 10M over the ballot kernel
   // (profiles/knn_kernel_history.md round 2). Any explicit
   // KAKVEDA_KNN_KERNEL selection other than 8pe disables it.
+  const bool use8pv3 = (ksel && std::string(ksel) == "8pv3") && N >= 65536 &&
+                       !emit_fallback;  // isolation A/B: v3 cold path
   const bool use8pe =
-      (ksel ? std::string(ksel) == "8pe" : k > 1) && N >= 65536 &&
-      !emit_fallback;
+      ((ksel ? std::string(ksel) == "8pe" : k > 1) || use8pv3) &&
+      N >= 65536 && !emit_fallback;
   const bool use8p =
       ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
       N >= 4096;
@@ -202,7 +204,18 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          samp_s.data_ptr<float>(),
                          (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
     }
-    hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
+    if (use8pv3)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<14>), grid, dim3(THREADS8),
+                         0, stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(),
+                         (unsigned long long*)nullptr, (float*)nullptr,
+                         (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), CAP);
+    else
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
                          stream.stream(), (const bf16_t*)queries.data_ptr(),
                          (const bf16_t*)corpus.data_ptr(),
                          pscore.data_ptr<float>(), pidx.data_ptr<int>(),
