@@ -185,6 +185,10 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
         {(long)B, CAP},
         torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
     auto ccount = torch::zeros({B}, opts_i);
+    // per-(block, wave) global emission stash (see kernels_impl.h EPI 9)
+    auto estash = torch::empty(
+        {(long)nchunks * row_tiles * 8 * ESTASH_STRIDE},
+        torch::TensorOptions().dtype(torch::kUInt8).device(queries.device()));
     // exact sample floor: merge the compact prepass partials into the
     // true top-8 of the whole sampled column set, publish its 8th as the
     // emission threshold (see publish_emission_floor). Block-per-row
@@ -213,7 +217,8 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          (unsigned*)rowthr.data_ptr<int>(),
                          (unsigned long long*)nullptr, (float*)nullptr,
                          (unsigned long long*)cand.data_ptr<int64_t>(),
-                         (unsigned*)ccount.data_ptr<int>(), CAP);
+                         (unsigned*)ccount.data_ptr<int>(), CAP,
+                         (char*)estash.data_ptr<uint8_t>());
     else
       hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
                          stream.stream(), (const bf16_t*)queries.data_ptr(),
@@ -223,7 +228,8 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          (unsigned*)rowthr.data_ptr<int>(),
                          (unsigned long long*)nullptr, (float*)nullptr,
                          (unsigned long long*)cand.data_ptr<int64_t>(),
-                         (unsigned*)ccount.data_ptr<int>(), CAP);
+                         (unsigned*)ccount.data_ptr<int>(), CAP,
+                         (char*)estash.data_ptr<uint8_t>());
     hipLaunchKernelGGL(emit_merge_topk, dim3(B), dim3(256), 0, stream.stream(),
                        (const unsigned long long*)cand.data_ptr<int64_t>(),
                        (const unsigned*)ccount.data_ptr<int>(),

@@ -685,6 +685,8 @@ constexpr int BM8 = 256;
 constexpr int BN8 = 256;
 constexpr int THREADS8 = 512;
 constexpr int HALF8 = 16384;  // one [128][64] bf16 half-image
+constexpr int EGCAP = 8;          // emission stash slots per (block, wave)
+constexpr int ESTASH_STRIDE = 64 + EGCAP * 1024;  // meta + slots, bytes
 
 
 
@@ -779,7 +781,8 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     unsigned* rowthr = nullptr, unsigned long long* stats = nullptr,
     float* __restrict__ slab = nullptr,
     unsigned long long* __restrict__ cand = nullptr,
-    unsigned* __restrict__ ccount = nullptr, long ccap = 0) {
+    unsigned* __restrict__ ccount = nullptr, long ccap = 0,
+    char* __restrict__ estash = nullptr) {
   __shared__ char smem[8 * HALF8 + 2 * BM8 * KMAX * 4 + 32 + BM8 * 4 + 32];
   char* const smem0 = smem;
   // buffer b in {0,1}: A half h at b*4*HALF8 + h*HALF8; B half h at +2*HALF8
@@ -1114,19 +1117,26 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
         }
       } else {
         if (__builtin_expect(qm32 != 0, 0)) {  // uniform cold path
-          // stash qualifying groups into the wave's 4 KiB slice of the
-          // dead last-window A-image: plain LDS stores only (see
-          // emit_stashed's comment for why no call may happen while the
-          // accumulators are live)
-          const int t_laste = j * nkt + nkt - 1;
-          char* stash = ahalf(t_laste & 1, 0) + wid * 4096;
+          // stash qualifying groups into this (block, wave)'s private
+          // slice of the GLOBAL emission scratch. An LDS stash in the
+          // "dead" last-window A-image corrupted later tiles' scores
+          // (isolated by the mode-14 A/B: v3 clean, LDS-stash dirty —
+          // the image interacts with the cross-tile staging pipeline in
+          // a way the stash-phase epilogues' barrier discipline tolerates
+          // but a late per-wave write does not); global scratch has no
+          // aliasing with the pipeline at all and needs no barrier.
+          char* stash =
+              estash +
+              ((size_t)(blockIdx.x + (size_t)gridDim.x * blockIdx.y) * 8 +
+               wid) *
+                  ESTASH_STRIDE;
           int ng = 0;
 #pragma unroll
           for (int m = 0; m < 8; ++m) {
 #pragma unroll
             for (int reg = 0; reg < 4; ++reg) {
               if (!(qm32 & (1u << (m * 4 + reg)))) continue;
-              if (ng < 3) {
+              if (ng < EGCAP) {
                 *(f32x4*)(stash + 64 + ng * 1024 + (size_t)lane * 16) =
                     f32x4{acc[m][0][reg], acc[m][1][reg], acc[m][2][reg],
                           acc[m][3][reg]};
@@ -1136,9 +1146,9 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
               ++ng;
             }
           }
-          if (__builtin_expect(ng > 3, 0)) {
-            // >3 qualifying groups in one wave-tile (possible only with
-            // degenerate floors): poison a row count -> host fallback
+          if (__builtin_expect(ng > EGCAP, 0)) {
+            // more qualifying groups than stash slots (degenerate floors
+            // only): poison a row count -> host fallback
             if (lane == 0)
               atomicAdd(&ccount[row0 < B ? row0 : B - 1],
                         (unsigned)(ccap + 1));
@@ -1147,18 +1157,6 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
                          colb, N, B, thr0, thr1);
           }
         }
-        // Every wave barriers so the next tile's A staging cannot land in
-        // a still-draining wave's stash slice (the slice spans another
-        // wave's staging quadrant). The vmcnt(0) is LOAD-BEARING: the
-        // drain's global atomics/stores are vmcnt-tracked on gfx9-family,
-        // and any left outstanding here would be counted by the next
-        // window's `vmcnt(4)` staging certification — which would then
-        // "prove" A/B tiles that have NOT landed and read uncertified
-        // LDS (garbage scores -> emission storms -> silent fallback;
-        // manifested box-dependently, the same counted-wait audit class
-        // as the round-1 chunk-tail race).
-        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_s_barrier();
       }
     } else {
       // ---- EPI_MODE 6/7 pre-check: one register ballot per (m,reg)
