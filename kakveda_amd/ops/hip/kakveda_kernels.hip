@@ -479,6 +479,72 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> floor_probe(
   return {samp_s, samp_i, rowthr};
 }
 
+torch::Tensor emit_counts_probe(torch::Tensor queries, torch::Tensor corpus,
+                                int64_t pregq) {
+  // Debug: run floors + the mode-9 emission kernel and return the raw
+  // per-row candidate counts (no merge, no fallback) for offline
+  // inspection of which rows overcount.
+  check_bf16_2d(queries, "queries");
+  check_bf16_2d(corpus, "corpus");
+  const int B = queries.size(0);
+  const int D = queries.size(1);
+  const int N = corpus.size(0);
+  const int preg = ((int)pregq) & ~7;
+  auto opts_f = torch::TensorOptions().dtype(torch::kFloat32).device(queries.device());
+  auto opts_i = torch::TensorOptions().dtype(torch::kInt32).device(queries.device());
+  auto rowthr = torch::empty({B}, opts_i);
+  auto ppre_s = torch::empty({(long)B * preg * KMAX}, opts_f);
+  auto ppre_i = torch::empty({(long)B * preg * KMAX}, opts_i);
+  auto samp_s = torch::empty({(long)B, (long)KMAX}, opts_f);
+  auto samp_i = torch::empty(
+      {(long)B, (long)KMAX},
+      torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(init_rowthr, dim3((B + 255) / 256), dim3(256), 0,
+                     stream.stream(), (unsigned*)rowthr.data_ptr<int>(), B);
+  constexpr int PRE_TILES = 8;
+  dim3 pgrid(preg, (B + BM - 1) / BM);
+  hipLaunchKernelGGL((cosine_topk_partial_t<11>), pgrid, dim3(THREADS), 0,
+                     stream.stream(), (const bf16_t*)queries.data_ptr(),
+                     (const bf16_t*)corpus.data_ptr(), ppre_s.data_ptr<float>(),
+                     ppre_i.data_ptr<int>(), B, N, D, PRE_TILES, preg,
+                     (unsigned*)rowthr.data_ptr<int>(),
+                     (unsigned long long*)nullptr);
+  hipLaunchKernelGGL(topk_merge, dim3(B), dim3(THREADS), 0, stream.stream(),
+                     ppre_s.data_ptr<float>(), ppre_i.data_ptr<int>(),
+                     samp_s.data_ptr<float>(),
+                     (long*)samp_i.data_ptr<int64_t>(), preg, KMAX);
+  hipLaunchKernelGGL(publish_emission_floor, dim3((B + 255) / 256), dim3(256),
+                     0, stream.stream(), samp_s.data_ptr<float>(),
+                     (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
+  const int row_tiles = (B + BM8 - 1) / BM8;
+  const int ntiles = (N + BN8 - 1) / BN8;
+  long want = ((long)ntiles * row_tiles + 511) / 512;
+  const int chunk_tiles = (int)std::max(4L, std::min(want, 128L));
+  const int nchunks = ((ntiles + chunk_tiles - 1) / chunk_tiles + 7) & ~7;
+  const long CAP = 32768;
+  auto pscore = torch::empty({(long)B * nchunks * KMAX}, opts_f);
+  auto pidx = torch::empty({(long)B * nchunks * KMAX}, opts_i);
+  auto cand = torch::empty(
+      {(long)B, CAP},
+      torch::TensorOptions().dtype(torch::kInt64).device(queries.device()));
+  auto ccount = torch::zeros({B}, opts_i);
+  auto estash = torch::empty(
+      {(long)nchunks * row_tiles * 8 * ESTASH_STRIDE},
+      torch::TensorOptions().dtype(torch::kUInt8).device(queries.device()));
+  dim3 grid(nchunks, row_tiles);
+  hipLaunchKernelGGL((cosine_topk_partial8p_t<9>), grid, dim3(THREADS8), 0,
+                     stream.stream(), (const bf16_t*)queries.data_ptr(),
+                     (const bf16_t*)corpus.data_ptr(), pscore.data_ptr<float>(),
+                     pidx.data_ptr<int>(), B, N, D, chunk_tiles, nchunks,
+                     (unsigned*)rowthr.data_ptr<int>(),
+                     (unsigned long long*)nullptr, (float*)nullptr,
+                     (unsigned long long*)cand.data_ptr<int64_t>(),
+                     (unsigned*)ccount.data_ptr<int>(), CAP,
+                     (char*)estash.data_ptr<uint8_t>());
+  return ccount;
+}
+
 double probe8p(torch::Tensor queries, torch::Tensor corpus, int64_t mode,
                int64_t iters) {
   // Timing probe for 8p epilogue isolation (profiles/knn_kernel_history):
@@ -554,5 +620,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "argmax-cosine assignment, LDS-resident centroids (C<=64)");
   m.def("probe8p", &probe8p, "8p epilogue-isolation timing probe");
   m.def("floor_probe", &floor_probe, "emission floor pipeline debug probe");
+  m.def("emit_counts_probe", &emit_counts_probe,
+        "emission per-row count debug probe");
   m.attr("KMAX") = KMAX;
 }
