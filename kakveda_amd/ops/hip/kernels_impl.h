@@ -1136,26 +1136,29 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 #pragma unroll
             for (int reg = 0; reg < 4; ++reg) {
               if (!(qm32 & (1u << (m * 4 + reg)))) continue;
-              if (ng < EGCAP) {
-                *(f32x4*)(stash + 64 + ng * 1024 + (size_t)lane * 16) =
-                    f32x4{acc[m][0][reg], acc[m][1][reg], acc[m][2][reg],
-                          acc[m][3][reg]};
-                if (lane == 0)
-                  ((volatile unsigned*)stash)[ng] = (unsigned)(m * 4 + reg);
+              if (__builtin_expect(ng == EGCAP, 0)) {
+                // stash full. Correlated query batches (e.g. the bench's
+                // template-generated signatures) make qualifying groups
+                // BURST on "hot" corpus tiles — dozens of groups at once
+                // — so overflow must drain-and-refill, not bail (a
+                // poison->host-fallback here fired ~600x/row-tile on the
+                // bench and doubled the step). This mid-sweep call keeps
+                // some accumulators live (cold-block spills only).
+                emit_stashed(cand, ccount, ccap, stash, ng,
+                             row0 + wr * 128, colb, N, B, thr0, thr1);
+                ng = 0;
               }
+              *(f32x4*)(stash + 64 + ng * 1024 + (size_t)lane * 16) =
+                  f32x4{acc[m][0][reg], acc[m][1][reg], acc[m][2][reg],
+                        acc[m][3][reg]};
+              if (lane == 0)
+                ((volatile unsigned*)stash)[ng] = (unsigned)(m * 4 + reg);
               ++ng;
             }
           }
-          if (__builtin_expect(ng > EGCAP, 0)) {
-            // more qualifying groups than stash slots (degenerate floors
-            // only): poison a row count -> host fallback
-            if (lane == 0)
-              atomicAdd(&ccount[row0 < B ? row0 : B - 1],
-                        (unsigned)(ccap + 1));
-          } else {
+          if (ng > 0)
             emit_stashed(cand, ccount, ccap, stash, ng, row0 + wr * 128,
                          colb, N, B, thr0, thr1);
-          }
         }
       }
     } else {
