@@ -479,8 +479,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> floor_probe(
   return {samp_s, samp_i, rowthr};
 }
 
-torch::Tensor emit_counts_probe(torch::Tensor queries, torch::Tensor corpus,
-                                int64_t pregq) {
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> emit_counts_probe(
+    torch::Tensor queries, torch::Tensor corpus, int64_t pregq) {
   // Debug: run floors + the mode-9 emission kernel and return the raw
   // per-row candidate counts (no merge, no fallback) for offline
   // inspection of which rows overcount.
@@ -542,7 +542,7 @@ torch::Tensor emit_counts_probe(torch::Tensor queries, torch::Tensor corpus,
                      (unsigned long long*)cand.data_ptr<int64_t>(),
                      (unsigned*)ccount.data_ptr<int>(), CAP,
                      (char*)estash.data_ptr<uint8_t>());
-  return ccount;
+  return {ccount, cand, rowthr};
 }
 
 double probe8p(torch::Tensor queries, torch::Tensor corpus, int64_t mode,
