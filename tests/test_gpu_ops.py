@@ -417,3 +417,37 @@ def test_emission_kernel_exact_vs_torch():
                        capture_output=True, text=True, timeout=600)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "OK8PE" in r.stdout
+
+
+def test_emission_correlated_queries_exact():
+    """Correlated (template-generated) query batches make qualifying
+    groups BURST on hot corpus tiles — the emission stash must
+    drain-and-refill, not bail (a fixed-capacity poison path fired
+    ~600x/row-tile on the bench and silently doubled step time via the
+    overflow fallback). Exactness on encoder-style correlated queries at
+    an emission-regime corpus size is the regression guard."""
+    from kakveda_amd import ops
+    from kakveda_amd.encoder.featurizer import featurize_batch
+    from kakveda_amd.encoder.model import TraceEncoder
+
+    dev = "cuda"
+    D, B, N = 768, 1024, 131072
+    enc = TraceEncoder(dim=D, hash_dim=1 << 16, seed=1234, device=dev)
+    texts = [
+        f"intent_tags:intent:citations_required | prompt_hint:synthetic probe "
+        f"{i} explain with sources | tools:t{i % 7} | env_keys:e2e,k{i % 5}"
+        for i in range(512)
+    ]
+    idx_np, w_np = featurize_batch(texts, hash_dim=enc.hash_dim, max_features=64)
+    fi = torch.from_numpy(idx_np).to(dev).repeat(2, 1)[:B].contiguous()
+    fw = torch.from_numpy(w_np).to(dev).repeat(2, 1)[:B].contiguous()
+    q = enc.encode_features(fi, fw).to(torch.bfloat16)
+    c = _rand_unit(N, D, seed=90)
+    scores, idx = ops.cosine_topk(q, c, 5)
+    torch.cuda.synchronize()
+    sims = q.float() @ c.float().t()
+    ref_s, _ = torch.topk(sims, 5, dim=1)
+    assert torch.allclose(scores, ref_s, atol=2e-2, rtol=1e-2)
+    gathered = sims.gather(1, idx)
+    assert torch.allclose(gathered, scores, atol=1e-4)
+    assert torch.allclose(gathered, ref_s, atol=1e-4)
