@@ -92,3 +92,37 @@ async def test_event_bus_topics(tmp_path):
     assert "trace.ingested" in topics
     assert "failure.detected" in topics
     await cluster.aclose()
+
+
+async def test_event_bus_durable_subscriptions(tmp_path, monkeypatch):
+    """KAKVEDA_BUS_DURABLE persists subscriptions across a bus restart
+    (the reference's in-memory bus loses them — a known limitation its
+    release notes call out; this is opt-in extra durability)."""
+    import httpx
+
+    from kakveda_amd.services.event_bus import create_app as bus_app
+
+    monkeypatch.setenv("KAKVEDA_BUS_DURABLE", str(tmp_path))
+    bus1 = bus_app()
+    async with httpx.AsyncClient(
+        transport=httpx.ASGITransport(app=bus1), base_url="http://bus"
+    ) as client:
+        r = await client.post(
+            "/subscribe",
+            json={"topic": "trace.ingested", "callback_url": "http://x/events"},
+        )
+        assert r.json()["subscribers"] == 1
+        # dedup does not double-append
+        await client.post(
+            "/subscribe",
+            json={"topic": "trace.ingested", "callback_url": "http://x/events"},
+        )
+    assert (tmp_path / "subscriptions.jsonl").read_text().count("http://x/events") == 1
+
+    # "restart": a fresh app instance reloads the subscription graph
+    bus2 = bus_app()
+    async with httpx.AsyncClient(
+        transport=httpx.ASGITransport(app=bus2), base_url="http://bus"
+    ) as client:
+        topics = (await client.get("/topics")).json()["topics"]
+        assert topics == {"trace.ingested": ["http://x/events"]}
