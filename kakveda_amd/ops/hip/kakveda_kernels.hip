@@ -49,8 +49,15 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // KAKVEDA_KNN_KERNEL selection other than 8pe disables it.
   const bool use8pv3 = (ksel && std::string(ksel) == "8pv3") && N >= 65536 &&
                        !emit_fallback;  // isolation A/B: v3 cold path
+  // Single-query serving against a 100M-scale corpus is the one measured
+  // corner where the ballot kernel still wins (p50 34.9 vs 41.5 ms at
+  // B=1/100M; emission wins B=1 at 1M and 10M): keep tiny batches on the
+  // ballot kernel at very large N.
+  const bool emit_small_batch_corner = B < 128 && N > 50000000;
   const bool use8pe =
-      ((ksel ? std::string(ksel) == "8pe" : k > 1) || use8pv3) &&
+      ((ksel ? std::string(ksel) == "8pe"
+             : (k > 1 && !emit_small_batch_corner)) ||
+       use8pv3) &&
       N >= 65536 && !emit_fallback;
   const bool use8p =
       ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
