@@ -451,3 +451,19 @@ def test_emission_correlated_queries_exact():
     gathered = sims.gather(1, idx)
     assert torch.allclose(gathered, scores, atol=1e-4)
     assert torch.allclose(gathered, ref_s, atol=1e-4)
+
+
+@pytest.mark.parametrize("D", [256, 1536])
+def test_emission_non_default_dims(D):
+    """The emission path at non-768 dims (D % 64 == 0): window counts,
+    prepass and floors must all generalise."""
+    from kakveda_amd import ops
+
+    q = _rand_unit(128, D, seed=95)
+    c = _rand_unit(131072, D, seed=96)
+    scores, idx = ops.cosine_topk(q, c, 5)
+    torch.cuda.synchronize()
+    sims = q.float() @ c.float().t()
+    ref_s, _ = torch.topk(sims, 5, dim=1)
+    assert torch.allclose(scores, ref_s, atol=2e-2, rtol=1e-2)
+    assert torch.allclose(sims.gather(1, idx), ref_s, atol=1e-4)
