@@ -146,3 +146,44 @@ def test_service_match_through_batcher(tmp_path):
     assert got and got[0]["failure_id"] == want[0].failure_id
     assert abs(got[0]["score"] - want[0].score) < 1e-6
     assert app.state.batcher.requests >= 1
+
+
+def test_concurrent_upserts_and_matches_stress(tmp_path):
+    """Thread hammer: continuous upserts of new identities while matches
+    run lock-free against count snapshots — no exceptions, matches always
+    reflect a consistent prefix, and the final state is fully queryable."""
+    eng = _engine(tmp_path)
+    stop = threading.Event()
+    errors = []
+
+    def inserter():
+        try:
+            for i in range(200):
+                eng.upsert_failure(
+                    "TIMEOUT", _sig(1000 + i), {"i": i}, app_id=f"app-{i % 5}"
+                )
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+        finally:
+            stop.set()
+
+    def matcher():
+        try:
+            while not stop.is_set():
+                m = eng.match(_sig(3))
+                assert m and m[0].failure_id  # identity map stays consistent
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=inserter)] + [
+        threading.Thread(target=matcher) for _ in range(3)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errors, errors
+    assert eng.store.count == 8 + 200
+    # every inserted identity is findable afterwards
+    m = eng.match(_sig(1199))
+    assert m and m[0].score >= 0.99
