@@ -49,16 +49,13 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // KAKVEDA_KNN_KERNEL selection other than 8pe disables it.
   const bool use8pv3 = (ksel && std::string(ksel) == "8pv3") && N >= 65536 &&
                        !emit_fallback;  // isolation A/B: v3 cold path
-  // Single-query serving against a 100M-scale corpus is the one measured
-  // corner where the ballot kernel still wins (p50 34.9 vs 41.5 ms at
-  // B=1/100M; emission wins B=1 at 1M and 10M): keep tiny batches on the
-  // ballot kernel at very large N.
-  const bool emit_small_batch_corner = B < 128 && N > 50000000;
   const bool use8pe =
-      ((ksel ? std::string(ksel) == "8pe"
-             : (k > 1 && !emit_small_batch_corner)) ||
-       use8pv3) &&
+      ((ksel ? std::string(ksel) == "8pe" : k > 1) || use8pv3) &&
       N >= 65536 && !emit_fallback;
+  // B <= 8 requests (single-query serving) skip the MFMA tile machinery
+  // entirely: the streaming smallb_emit_kernel reads the corpus once at
+  // full bandwidth with the same emission floors/merge
+  const bool use_smallb = use8pe && B <= 8;
   const bool use8p =
       ((ksel && std::string(ksel) == "8p") || use8pbl || use8pq || use8pe) &&
       N >= 4096;
@@ -215,7 +212,16 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          samp_s.data_ptr<float>(),
                          (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
     }
-    if (use8pv3)
+    if (use_smallb) {
+      const long nblk = std::min((long)((N + 255) / 256), 4096L);
+      hipLaunchKernelGGL(smallb_emit_kernel, dim3((int)nblk), dim3(256),
+                         (size_t)B * D * 2, stream.stream(),
+                         (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(), B, (long)N, D,
+                         (const unsigned*)rowthr.data_ptr<int>(),
+                         (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), CAP);
+    } else if (use8pv3)
       hipLaunchKernelGGL((cosine_topk_partial8p_t<14>), grid, dim3(THREADS8),
                          0, stream.stream(), (const bf16_t*)queries.data_ptr(),
                          (const bf16_t*)corpus.data_ptr(),

@@ -467,3 +467,25 @@ def test_emission_non_default_dims(D):
     ref_s, _ = torch.topk(sims, 5, dim=1)
     assert torch.allclose(scores, ref_s, atol=2e-2, rtol=1e-2)
     assert torch.allclose(sims.gather(1, idx), ref_s, atol=1e-4)
+
+
+@pytest.mark.parametrize("B", [1, 3, 8])
+def test_smallb_streaming_search_exact(B):
+    """B <= 8 requests route to the streaming small-batch emission kernel
+    (no MFMA tile padding); must stay exact vs fp32 torch."""
+    from kakveda_amd import ops
+
+    q = _rand_unit(B, 768, seed=100 + B)
+    c = _rand_unit(300000, 768, seed=101)
+    scores, idx = ops.cosine_topk(q, c, 5)
+    torch.cuda.synchronize()
+    sims = q.float() @ c.float().t()
+    ref_s, _ = torch.topk(sims, 5, dim=1)
+    assert torch.allclose(scores, ref_s, atol=2e-2, rtol=1e-2)
+    assert torch.allclose(sims.gather(1, idx), ref_s, atol=1e-4)
+    # valid_n prefix respected on the streaming path too
+    scores2, idx2 = ops.cosine_topk(q, c, 5, valid_n=123457)
+    torch.cuda.synchronize()
+    assert (idx2 < 123457).all()
+    ref2, _ = torch.topk(sims[:, :123457], 5, dim=1)
+    assert torch.allclose(scores2, ref2, atol=2e-2, rtol=1e-2)
