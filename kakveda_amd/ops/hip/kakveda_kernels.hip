@@ -213,7 +213,7 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
     }
     if (use_smallb) {
-      const long nblk = std::min((long)((N + 255) / 256), 4096L);
+      const long nblk = std::min((long)((N + 255) / 256), 8192L);
       hipLaunchKernelGGL(smallb_emit_kernel, dim3((int)nblk), dim3(256),
                          (size_t)B * D * 2, stream.stream(),
                          (const bf16_t*)queries.data_ptr(),

@@ -1486,26 +1486,39 @@ __global__ __launch_bounds__(256) void smallb_emit_kernel(
   if (threadIdx.x < B) fl[threadIdx.x] = dec_f32(rowthr[threadIdx.x]);
   __syncthreads();
 
-  const int nj = D / 8;
+  const int nj = D / 8;  // D % 64 == 0 -> nj % 8 == 0
   for (long r = (long)blockIdx.x * 256 + threadIdx.x; r < N;
        r += (long)gridDim.x * 256) {
-    float acc[8];
+    // 4 independent partial accumulators per query: a single running
+    // accumulator serializes a 768-deep FMA dependency chain per row
+    // (measured 2.5x slower than the padded MFMA path); 4 chains + the
+    // 4-vector load batch restore ILP and memory-level parallelism.
+    f32x4 accv[8];
 #pragma unroll
-    for (int b = 0; b < 8; ++b) acc[b] = 0.f;
+    for (int b = 0; b < 8; ++b) accv[b] = f32x4{0.f, 0.f, 0.f, 0.f};
     const bf16x8* row = (const bf16x8*)(C + r * D);
-#pragma unroll 4
-    for (int j = 0; j < nj; ++j) {
-      const bf16x8 cv = row[j];
+    for (int j = 0; j < nj; j += 4) {
+      const bf16x8 cv0 = row[j], cv1 = row[j + 1];
+      const bf16x8 cv2 = row[j + 2], cv3 = row[j + 3];
 #pragma unroll
       for (int b = 0; b < 8; ++b) {
         if (b < B) {
-          const bf16x8 qv = *(const bf16x8*)(qmem + ((size_t)b * D + j * 8) * 2);
+          const bf16x8* qb = (const bf16x8*)(qmem + (size_t)b * D * 2) + j;
+          const bf16x8 q0 = qb[0], q1 = qb[1], q2 = qb[2], q3 = qb[3];
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            acc[b] += (float)cv[e] * (float)qv[e];
+          for (int e = 0; e < 8; ++e) {
+            accv[b][0] += (float)cv0[e] * (float)q0[e];
+            accv[b][1] += (float)cv1[e] * (float)q1[e];
+            accv[b][2] += (float)cv2[e] * (float)q2[e];
+            accv[b][3] += (float)cv3[e] * (float)q3[e];
+          }
         }
       }
     }
+    float acc[8];
+#pragma unroll
+    for (int b = 0; b < 8; ++b)
+      acc[b] = (accv[b][0] + accv[b][1]) + (accv[b][2] + accv[b][3]);
 #pragma unroll
     for (int b = 0; b < 8; ++b) {
       if (b < B && acc[b] >= fl[b]) {
