@@ -1493,24 +1493,25 @@ __global__ __launch_bounds__(256) void smallb_emit_kernel(
     // accumulator serializes a 768-deep FMA dependency chain per row
     // (measured 2.5x slower than the padded MFMA path); 4 chains + the
     // 4-vector load batch restore ILP and memory-level parallelism.
-    f32x4 accv[8];
+    f32x4 accv[8][2];
 #pragma unroll
-    for (int b = 0; b < 8; ++b) accv[b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < 8; ++b)
+      accv[b][0] = accv[b][1] = f32x4{0.f, 0.f, 0.f, 0.f};
     const bf16x8* row = (const bf16x8*)(C + r * D);
-    for (int j = 0; j < nj; j += 4) {
-      const bf16x8 cv0 = row[j], cv1 = row[j + 1];
-      const bf16x8 cv2 = row[j + 2], cv3 = row[j + 3];
+    for (int j = 0; j < nj; j += 8) {
+      bf16x8 cv[8];
+#pragma unroll
+      for (int v = 0; v < 8; ++v) cv[v] = row[j + v];
 #pragma unroll
       for (int b = 0; b < 8; ++b) {
         if (b < B) {
           const bf16x8* qb = (const bf16x8*)(qmem + (size_t)b * D * 2) + j;
-          const bf16x8 q0 = qb[0], q1 = qb[1], q2 = qb[2], q3 = qb[3];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            accv[b][0] += (float)cv0[e] * (float)q0[e];
-            accv[b][1] += (float)cv1[e] * (float)q1[e];
-            accv[b][2] += (float)cv2[e] * (float)q2[e];
-            accv[b][3] += (float)cv3[e] * (float)q3[e];
+          for (int v = 0; v < 8; ++v) {
+            const bf16x8 qv = qb[v];
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              accv[b][v >> 2][v & 3] += (float)cv[v][e] * (float)qv[e];
           }
         }
       }
@@ -1518,7 +1519,10 @@ __global__ __launch_bounds__(256) void smallb_emit_kernel(
     float acc[8];
 #pragma unroll
     for (int b = 0; b < 8; ++b)
-      acc[b] = (accv[b][0] + accv[b][1]) + (accv[b][2] + accv[b][3]);
+      acc[b] = ((accv[b][0][0] + accv[b][0][1]) +
+                (accv[b][0][2] + accv[b][0][3])) +
+               ((accv[b][1][0] + accv[b][1][1]) +
+                (accv[b][1][2] + accv[b][1][3]));
 #pragma unroll
     for (int b = 0; b < 8; ++b) {
       if (b < B && acc[b] >= fl[b]) {
