@@ -370,3 +370,24 @@ async def test_bootstrap_password_repair_gated(tmp_path, monkeypatch):
         admin = s.query(dbm.User).filter_by(email="admin@kakveda.local").first()
         assert verify_password("admin123", admin.password_hash)
     await cluster.aclose()
+
+
+async def test_theme_assets_served(tmp_path):
+    """The theme ships as real static assets (stylesheet, logo, JS) and
+    pages link them (round-2 presentation pass)."""
+    cluster = await _cluster(tmp_path)
+    async with _client(cluster) as client:
+        css = await client.get("/static/style.css")
+        assert css.status_code == 200
+        assert "text/css" in css.headers["content-type"]
+        assert ".topbar" in css.text and ".tile" in css.text and ".chart" in css.text
+        logo = await client.get("/static/logo.svg")
+        assert logo.status_code == 200 and "svg" in logo.headers["content-type"]
+        js = await client.get("/static/app.js")
+        assert js.status_code == 200 and "pg-form" in js.text
+        login = await client.get("/login")
+        assert '/static/style.css' in login.text and 'class="brand"' in login.text
+        await _login(client)
+        home = await client.get("/")
+        assert 'class="tiles"' in home.text and 'class="topbar"' in home.text
+    await cluster.aclose()
