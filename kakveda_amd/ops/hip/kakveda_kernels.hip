@@ -49,8 +49,13 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
   // KAKVEDA_KNN_KERNEL selection other than 8pe disables it.
   const bool use8pv3 = (ksel && std::string(ksel) == "8pv3") && N >= 65536 &&
                        !emit_fallback;  // isolation A/B: v3 cold path
+  // 8pe2 (EPI_MODE 15): emission epilogue with block-lifetime threshold
+  // vectors — the per-tile hot sweep drops its 32 ds_bpermute + 32
+  // ballots for a pure-VALU running max + one ballot (kernels_impl.h)
+  const bool use8pv2 = (ksel && std::string(ksel) == "8pe2") && N >= 65536 &&
+                       !emit_fallback;
   const bool use8pe =
-      ((ksel ? std::string(ksel) == "8pe" : k > 1) || use8pv3) &&
+      ((ksel ? std::string(ksel) == "8pe" : k > 1) || use8pv3 || use8pv2) &&
       N >= 65536 && !emit_fallback;
   // B <= 8 requests (single-query serving) skip the MFMA tile machinery
   // entirely: the streaming smallb_emit_kernel reads the corpus once at
@@ -223,6 +228,17 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          (unsigned*)ccount.data_ptr<int>(), CAP);
     } else if (use8pv3)
       hipLaunchKernelGGL((cosine_topk_partial8p_t<14>), grid, dim3(THREADS8),
+                         0, stream.stream(), (const bf16_t*)queries.data_ptr(),
+                         (const bf16_t*)corpus.data_ptr(),
+                         pscore.data_ptr<float>(), pidx.data_ptr<int>(),
+                         B, N, D, chunk_tiles, nchunks,
+                         (unsigned*)rowthr.data_ptr<int>(),
+                         (unsigned long long*)nullptr, (float*)nullptr,
+                         (unsigned long long*)cand.data_ptr<int64_t>(),
+                         (unsigned*)ccount.data_ptr<int>(), CAP,
+                         (char*)estash.data_ptr<uint8_t>());
+    else if (use8pv2)
+      hipLaunchKernelGGL((cosine_topk_partial8p_t<15>), grid, dim3(THREADS8),
                          0, stream.stream(), (const bf16_t*)queries.data_ptr(),
                          (const bf16_t*)corpus.data_ptr(),
                          pscore.data_ptr<float>(), pidx.data_ptr<int>(),

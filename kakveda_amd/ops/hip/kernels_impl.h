@@ -888,6 +888,49 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 
   const int total_windows = tiles_here * nkt;
 
+  // EPI_MODE 15: block-lifetime per-lane threshold VECTORS. Under the
+  // emission main launch rowthr is static (prepass-published floors; mode
+  // 9/15 never write it), so the 32 per-(m,reg) thresholds each lane
+  // compares against are constants for the whole block. Mode 9 re-derives
+  // them with 32 ds_bpermute + 32 vcc-serialised ballots per TILE; here
+  // they are shuffled ONCE into 16 registers as bf16 PAIRS (a full-f32
+  // tv[32] spilled 32 B/lane at the 256-VGPR/2-wave cap), rounded toward
+  // -inf so the packed floor is <= the exact one. The per-tile hot sweep
+  // becomes a pure-VALU running max-diff with a single ballot.
+  // Exactness: qualification vs the rounded-DOWN floor is a superset of
+  // qualification vs the exact floor, and a hot-sweep positive only opens
+  // the cold path, whose per-lane compares in emit_stashed re-check the
+  // EXACT thr0/thr1 before any candidate is stored.
+  unsigned tvp[16];
+  if constexpr (EPI_MODE == 15) {
+    const int r0g = row0 + wr * 128 + lane;
+    const float t0b =
+        (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
+    const float t1b = (r0g + 64 < B)
+                          ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
+                          : 1e38f;
+    // upper-16 f32 bits rounded toward -inf (truncation rounds toward
+    // zero, i.e. UP for negatives -> push one ulp down there; a
+    // most-negative-finite pushed to 0xff80 = -inf still compares
+    // conservatively)
+    auto bfloor = [](float f) -> unsigned {
+      const unsigned b = __float_as_uint(f);
+      unsigned hi = b >> 16;
+      if ((b & 0xffffu) && (b >> 31)) ++hi;
+      return hi & 0xffffu;
+    };
+#pragma unroll
+    for (int m = 0; m < 8; ++m)
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int rl0 = m * 16 + g * 4 + 2 * h;
+        const unsigned lo = bfloor(__shfl(m >= 4 ? t1b : t0b, rl0 & 63, 64));
+        const unsigned hi =
+            bfloor(__shfl(m >= 4 ? t1b : t0b, (rl0 + 1) & 63, 64));
+        tvp[m * 2 + h] = lo | (hi << 16);
+      }
+  }
+
   // slab-deferred epilogue state (EPI_MODE 5): this block's 256 KiB slab.
   // Thread tid<256 owns list row tid; its warm threshold is cached in a
   // register per tile. The drain is array-free (slot 0 of a list always
@@ -955,7 +998,7 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
     // rows (rows beyond B get +inf so clamped-row garbage never flags)
     float thr0 = NEG_INF, thr1 = NEG_INF;
     if constexpr (EPI_MODE == 6 || EPI_MODE == 7 || EPI_MODE == 9 ||
-                  EPI_MODE == 12 || EPI_MODE == 14) {
+                  EPI_MODE == 12 || EPI_MODE == 14 || EPI_MODE == 15) {
       const int r0g = row0 + wr * 128 + lane;
       thr0 = (r0g < B) ? (rowthr ? dec_f32(rowthr[r0g]) : NEG_INF) : 1e38f;
       thr1 = (r0g + 64 < B) ? (rowthr ? dec_f32(rowthr[r0g + 64]) : NEG_INF)
@@ -1056,7 +1099,8 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
 #pragma unroll
         for (int n = 0; n < 4; ++n)
           asm volatile("" ::"v"(acc[m][n]));
-    } else if constexpr (EPI_MODE == 9 || EPI_MODE == 12 || EPI_MODE == 14) {
+    } else if constexpr (EPI_MODE == 9 || EPI_MODE == 12 || EPI_MODE == 14 ||
+                         EPI_MODE == 15) {
       // ---- threshold-emission epilogue (round 2): no lists, no stash
       // phases, no barriers — the 1017 TF GEMM core's full accumulator
       // sweep is a register compare + rare global append. Exactness: the
@@ -1078,16 +1122,48 @@ __global__ __launch_bounds__(THREADS8, 2) void cosine_topk_partial8p_t(
       // calls; its spills live in that cold block only.
       const int colb = col0 + wc * 64 + cl;
       unsigned qm32 = 0;
+      if constexpr (EPI_MODE == 15) {
+        // mode 15 hot sweep: no shuffles (tv precomputed per block), no
+        // per-group ballots — one running max of (gmax - thr) and ONE
+        // wave ballot. The per-group qualification bits are only needed
+        // on the rare cold path, where they are re-derived with the
+        // ballots mode 9 pays unconditionally.
+        auto thr_lo = [&](int m, int reg) -> float {
+          const unsigned pk = tvp[m * 2 + (reg >> 1)];
+          return __uint_as_float((reg & 1) ? (pk & 0xffff0000u) : (pk << 16));
+        };
+        float run = NEG_INF;
 #pragma unroll
-      for (int m = 0; m < 8; ++m) {
+        for (int m = 0; m < 8; ++m)
 #pragma unroll
-        for (int reg = 0; reg < 4; ++reg) {
-          const int rl = m * 16 + g * 4 + reg;  // 0..127 within this half
-          const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
-          const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
-                                   fmaxf(acc[m][2][reg], acc[m][3][reg]));
-          if (__ballot(gmax >= thr))  // uniform
-            qm32 |= 1u << (m * 4 + reg);
+          for (int reg = 0; reg < 4; ++reg) {
+            const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                                     fmaxf(acc[m][2][reg], acc[m][3][reg]));
+            run = fmaxf(run, gmax - thr_lo(m, reg));
+          }
+        if (__builtin_expect(__ballot(run >= 0.f) != 0, 0)) {
+#pragma unroll
+          for (int m = 0; m < 8; ++m)
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+              const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                                       fmaxf(acc[m][2][reg], acc[m][3][reg]));
+              if (__ballot(gmax >= thr_lo(m, reg)))
+                qm32 |= 1u << (m * 4 + reg);
+            }
+        }
+      } else {
+#pragma unroll
+        for (int m = 0; m < 8; ++m) {
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) {
+            const int rl = m * 16 + g * 4 + reg;  // 0..127 within this half
+            const float thr = __shfl(m >= 4 ? thr1 : thr0, rl & 63, 64);
+            const float gmax = fmaxf(fmaxf(acc[m][0][reg], acc[m][1][reg]),
+                                     fmaxf(acc[m][2][reg], acc[m][3][reg]));
+            if (__ballot(gmax >= thr))  // uniform
+              qm32 |= 1u << (m * 4 + reg);
+          }
         }
       }
       if constexpr (EPI_MODE == 12) {

@@ -244,7 +244,7 @@ def test_k1_argmax_fast_path():
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("variant", ["eager", "fast", "rege", "8pbl", "8pe"])
+@pytest.mark.parametrize("variant", ["eager", "fast", "rege", "8pbl", "8pe", "8pe2"])
 def test_kernel_variants_match_default(variant):
     """Every KAKVEDA_KNN_KERNEL variant must produce the same top-k
     scores as the default ballot kernel (env is read once per process,
@@ -255,7 +255,7 @@ def test_kernel_variants_match_default(variant):
     import subprocess
     import sys
 
-    n = 131072 if variant == "8pe" else 8192
+    n = 131072 if variant in ("8pe", "8pe2") else 8192
     code = (
         "import torch\n"
         "from kakveda_amd import ops\n"
