@@ -218,14 +218,26 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
     }
     if (use_smallb) {
+      // KAKVEDA_SMALLB=4 selects the 8-lanes-per-row remap (v4); default
+      // stays the per-lane-row streaming kernel (v2/v3)
+      static const char* sbenv = std::getenv("KAKVEDA_SMALLB");
       const long nblk = std::min((long)((N + 255) / 256), 8192L);
-      hipLaunchKernelGGL(smallb_emit_kernel, dim3((int)nblk), dim3(256),
-                         (size_t)B * D * 2, stream.stream(),
-                         (const bf16_t*)queries.data_ptr(),
-                         (const bf16_t*)corpus.data_ptr(), B, (long)N, D,
-                         (const unsigned*)rowthr.data_ptr<int>(),
-                         (unsigned long long*)cand.data_ptr<int64_t>(),
-                         (unsigned*)ccount.data_ptr<int>(), CAP);
+      if (sbenv && sbenv[0] == '4')
+        hipLaunchKernelGGL(smallb_emit_kernel_v4, dim3((int)nblk), dim3(256),
+                           (size_t)B * D * 2, stream.stream(),
+                           (const bf16_t*)queries.data_ptr(),
+                           (const bf16_t*)corpus.data_ptr(), B, (long)N, D,
+                           (const unsigned*)rowthr.data_ptr<int>(),
+                           (unsigned long long*)cand.data_ptr<int64_t>(),
+                           (unsigned*)ccount.data_ptr<int>(), CAP);
+      else
+        hipLaunchKernelGGL(smallb_emit_kernel, dim3((int)nblk), dim3(256),
+                           (size_t)B * D * 2, stream.stream(),
+                           (const bf16_t*)queries.data_ptr(),
+                           (const bf16_t*)corpus.data_ptr(), B, (long)N, D,
+                           (const unsigned*)rowthr.data_ptr<int>(),
+                           (unsigned long long*)cand.data_ptr<int64_t>(),
+                           (unsigned*)ccount.data_ptr<int>(), CAP);
     } else if (use8pv3)
       hipLaunchKernelGGL((cosine_topk_partial8p_t<14>), grid, dim3(THREADS8),
                          0, stream.stream(), (const bf16_t*)queries.data_ptr(),

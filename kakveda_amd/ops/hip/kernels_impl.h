@@ -1612,6 +1612,93 @@ __global__ __launch_bounds__(256) void smallb_emit_kernel(
   }
 }
 
+// v4 lane remap (KAKVEDA_SMALLB=4): one wave covers 8 CONSECUTIVE rows
+// with 8 lanes per row, so a single load instruction touches 8 FULL
+// 128-byte lines. v2's per-lane-row mapping touches 64 lines per
+// instruction, 16 B each — a line must survive 7 more instructions in
+// L1 to be fully consumed, and with 8 resident waves the combined
+// window exceeds L1, so lines get refetched from L2 (the suspected
+// ~4.7-of-8 TB/s efficiency gap at 100M). Scores are reduced across the
+// 8 chunk lanes with 3 shfl_xor steps at row end; per-lane f32x4
+// sub-chains keep the FMA dependency depth at 2 per segment (the v1
+// lesson). Exactness: identical emission contract to v2 — every score
+// >= the shared prepass floor is emitted; tail rows are clamped for the
+// load and guarded at emission.
+__global__ __launch_bounds__(256) void smallb_emit_kernel_v4(
+    const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C, int B,
+    long N, int D, const unsigned* __restrict__ rowthr,
+    unsigned long long* __restrict__ cand, unsigned* __restrict__ ccount,
+    long ccap) {
+  extern __shared__ char qmem[];  // [B][D] bf16 queries
+  __shared__ float fl[8];
+  for (int i = threadIdx.x; i < B * (D / 8); i += 256)
+    ((bf16x8*)qmem)[i] = ((const bf16x8*)Q)[i];
+  if (threadIdx.x < B) fl[threadIdx.x] = dec_f32(rowthr[threadIdx.x]);
+  __syncthreads();
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int rr = lane >> 3;  // row within this wave's 8
+  const int ch = lane & 7;   // 16-byte chunk within a 128-B segment
+  const int nseg = D / 64;   // 128-B segments per row (D % 64 == 0)
+  for (long r0 = (long)blockIdx.x * 32 + wid * 8; r0 < N;
+       r0 += (long)gridDim.x * 32) {
+    const long r = r0 + rr < N ? r0 + rr : N - 1;  // clamp tail loads
+    const bf16x8* row = (const bf16x8*)(C + r * D);
+    f32x4 accv[8];
+#pragma unroll
+    for (int b = 0; b < 8; ++b) accv[b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    int j = 0;
+    for (; j + 4 <= nseg; j += 4) {
+      bf16x8 cv[4];
+#pragma unroll
+      for (int v = 0; v < 4; ++v) cv[v] = row[(j + v) * 8 + ch];
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        if (b < B) {
+          const bf16x8* qb = (const bf16x8*)(qmem + (size_t)b * D * 2);
+#pragma unroll
+          for (int v = 0; v < 4; ++v) {
+            const bf16x8 qv = qb[(j + v) * 8 + ch];
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              accv[b][e & 3] += (float)cv[v][e] * (float)qv[e];
+          }
+        }
+      }
+    }
+    for (; j < nseg; ++j) {  // D % 256 != 0 tail segments
+      const bf16x8 cv = row[j * 8 + ch];
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        if (b < B) {
+          const bf16x8 qv =
+              ((const bf16x8*)(qmem + (size_t)b * D * 2))[j * 8 + ch];
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            accv[b][e & 3] += (float)cv[e] * (float)qv[e];
+        }
+      }
+    }
+#pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      if (b < B) {
+        float s = (accv[b][0] + accv[b][1]) + (accv[b][2] + accv[b][3]);
+        s += __shfl_xor(s, 1, 64);
+        s += __shfl_xor(s, 2, 64);
+        s += __shfl_xor(s, 4, 64);  // the row group's 8 lanes now agree
+        if (ch == 0 && r0 + rr < N && s >= fl[b]) {
+          const unsigned pos = atomicAdd(&ccount[b], 1u);
+          if (pos < (unsigned)ccap)
+            cand[(size_t)b * ccap + pos] =
+                ((unsigned long long)enc_f32(s) << 32) |
+                (unsigned)(0x7fffffff - (int)(r0 + rr));
+        }
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Companion merge for the emission epilogue (8p EPI_MODE 9): exact top-k
 // per row over the emitted (score, col) candidates. grid = B blocks x 256

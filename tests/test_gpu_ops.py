@@ -489,3 +489,42 @@ def test_smallb_streaming_search_exact(B):
     assert (idx2 < 123457).all()
     ref2, _ = torch.topk(sims[:, :123457], 5, dim=1)
     assert torch.allclose(scores2, ref2, atol=2e-2, rtol=1e-2)
+
+
+@pytest.mark.gpu
+def test_smallb_v4_matches_default():
+    """KAKVEDA_SMALLB=4 (8-lanes-per-row remap) must produce the same
+    top-k as the default streaming kernel (env read once per process, so
+    the variant runs in a subprocess)."""
+    import os
+    import subprocess
+    import sys
+
+    code = (
+        "import torch\n"
+        "from kakveda_amd import ops\n"
+        "q = torch.randn(3, 768, generator=torch.Generator(device='cuda')"
+        ".manual_seed(7), device='cuda')\n"
+        "q = (q / q.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
+        "c = torch.randn(300000, 768, generator=torch.Generator(device='cuda')"
+        ".manual_seed(8), device='cuda')\n"
+        "c = (c / c.norm(dim=-1, keepdim=True)).to(torch.bfloat16)\n"
+        "s, i = ops.cosine_topk(q, c, 5)\n"
+        "s2, i2 = ops.cosine_topk(q, c, 5, valid_n=123457)\n"
+        "torch.cuda.synchronize()\n"
+        "assert (i2 < 123457).all()\n"
+        "print('CSUM', float(s.double().sum()) + float(s2.double().sum()))\n"
+    )
+    outs = {}
+    for sel in (None, "4"):
+        env = dict(os.environ)
+        env.pop("KAKVEDA_SMALLB", None)
+        if sel:
+            env["KAKVEDA_SMALLB"] = sel
+        r = subprocess.run(
+            [sys.executable, "-c", code], env=env, capture_output=True,
+            text=True, timeout=300,
+        )
+        assert r.returncode == 0, r.stderr[-1500:]
+        outs[sel] = float(r.stdout.split("CSUM")[1].strip())
+    assert abs(outs[None] - outs["4"]) < 1e-2, outs
