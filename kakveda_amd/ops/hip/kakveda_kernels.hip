@@ -218,11 +218,13 @@ std::tuple<torch::Tensor, torch::Tensor> cosine_topk(
                          (unsigned*)rowthr.data_ptr<int>(), B, KMAX - 1);
     }
     if (use_smallb) {
-      // KAKVEDA_SMALLB=4 selects the 8-lanes-per-row remap (v4); default
-      // stays the per-lane-row streaming kernel (v2/v3)
+      // Default: the v4 8-lanes-per-row remap (same-box A/B at 10M:
+      // +1.5% at B=1, +13.6% at B=4, exact parity — see
+      // profiles/knn_kernel_history.md). KAKVEDA_SMALLB=2 reverts to the
+      // per-lane-row streaming kernel (v2/v3).
       static const char* sbenv = std::getenv("KAKVEDA_SMALLB");
       const long nblk = std::min((long)((N + 255) / 256), 8192L);
-      if (sbenv && sbenv[0] == '4')
+      if (!(sbenv && sbenv[0] == '2'))
         hipLaunchKernelGGL(smallb_emit_kernel_v4, dim3((int)nblk), dim3(256),
                            (size_t)B * D * 2, stream.stream(),
                            (const bf16_t*)queries.data_ptr(),

@@ -493,9 +493,9 @@ def test_smallb_streaming_search_exact(B):
 
 @pytest.mark.gpu
 def test_smallb_v4_matches_default():
-    """KAKVEDA_SMALLB=4 (8-lanes-per-row remap) must produce the same
-    top-k as the default streaming kernel (env read once per process, so
-    the variant runs in a subprocess)."""
+    """The default v4 (8-lanes-per-row remap) and the legacy v2
+    per-lane-row kernel (KAKVEDA_SMALLB=2) must produce the same top-k
+    (env read once per process, so the variant runs in a subprocess)."""
     import os
     import subprocess
     import sys
@@ -516,7 +516,7 @@ def test_smallb_v4_matches_default():
         "print('CSUM', float(s.double().sum()) + float(s2.double().sum()))\n"
     )
     outs = {}
-    for sel in (None, "4"):
+    for sel in (None, "2"):
         env = dict(os.environ)
         env.pop("KAKVEDA_SMALLB", None)
         if sel:
@@ -527,4 +527,4 @@ def test_smallb_v4_matches_default():
         )
         assert r.returncode == 0, r.stderr[-1500:]
         outs[sel] = float(r.stdout.split("CSUM")[1].strip())
-    assert abs(outs[None] - outs["4"]) < 1e-2, outs
+    assert abs(outs[None] - outs["2"]) < 1e-2, outs

@@ -48,11 +48,12 @@ for B in (1, 4):
 
 for N in NS:
     res = {}
+    # after the default flip: "v2" forces the legacy kernel, "4" is default
     for sel in ("v2", "4"):
         env = dict(os.environ)
         env.pop("KAKVEDA_SMALLB", None)
-        if sel == "4":
-            env["KAKVEDA_SMALLB"] = "4"
+        if sel == "v2":
+            env["KAKVEDA_SMALLB"] = "2"
         r = subprocess.run([sys.executable, "-c", CODE, str(N)], env=env,
                            capture_output=True, text=True, timeout=900)
         if r.returncode != 0:
