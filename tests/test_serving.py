@@ -187,3 +187,27 @@ def test_concurrent_upserts_and_matches_stress(tmp_path):
     # every inserted identity is findable afterwards
     m = eng.match(_sig(1199))
     assert m and m[0].score >= 0.99
+
+
+def test_http_bench_harness_runs():
+    """The multi-process HTTP serving benchmark (real uvicorn over TCP +
+    separate client processes) completes and reports sane numbers on the
+    CPU fallback path with a tiny corpus."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "benchmarks", "serve_http_bench.py"),
+         "--entries", "2000", "--seconds", "2", "--procs", "1",
+         "--conns", "4", "--workers", "1", "--port", "8217"],
+        capture_output=True, text=True, timeout=240, cwd=root,
+    )
+    assert r.returncode == 0, r.stderr[-1500:]
+    out = json.loads(r.stdout.splitlines()[-1])
+    assert out["metric"] == "serve_warn_http"
+    assert out["requests"] > 0 and out["value"] > 0
+    assert out["matched"] > 0  # the seeded failure matches the demo prompt
+    assert out["p99_ms"] >= out["p50_ms"] > 0
