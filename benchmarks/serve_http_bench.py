@@ -43,33 +43,44 @@ def client_main(url: str, seconds: float, conns: int) -> None:
     async def run() -> None:
         lats: list[float] = []
         matched = 0
+        errors = 0
         async with httpx.AsyncClient(
             base_url=url,
-            timeout=30.0,
+            timeout=120.0,
             limits=httpx.Limits(max_connections=conns,
                                 max_keepalive_connections=conns),
         ) as cli:
-            # per-connection warmup
-            for _ in range(conns):
-                await cli.post("/warn", json=BODY)
+            # concurrent warmup (sequential per-connection warmup at high
+            # fan-in takes minutes against a saturated server)
+            await asyncio.gather(
+                *(cli.post("/warn", json=BODY) for _ in range(min(conns, 32))),
+                return_exceptions=True)
             stop_at = time.perf_counter() + seconds
 
             async def loop() -> None:
-                nonlocal matched
+                nonlocal matched, errors
                 while time.perf_counter() < stop_at:
                     t0 = time.perf_counter()
-                    r = await cli.post("/warn", json=BODY)
-                    lats.append(time.perf_counter() - t0)
-                    if r.status_code == 200 and r.json().get("references"):
-                        matched += 1
+                    try:
+                        r = await cli.post("/warn", json=BODY)
+                        lats.append(time.perf_counter() - t0)
+                        if r.status_code == 200 and r.json().get("references"):
+                            matched += 1
+                        elif r.status_code != 200:
+                            errors += 1
+                    except Exception:
+                        errors += 1
 
             t0 = time.perf_counter()
-            await asyncio.gather(*(loop() for _ in range(conns)))
+            res = await asyncio.gather(*(loop() for _ in range(conns)),
+                                       return_exceptions=True)
+            errors += sum(1 for x in res if isinstance(x, Exception))
             wall = time.perf_counter() - t0
         print(json.dumps({
             "count": len(lats),
             "wall": wall,
             "matched": matched,
+            "errors": errors,
             "lats_ms": [round(x * 1000, 3) for x in lats],
         }))
 
@@ -136,13 +147,18 @@ def main() -> int:
             for _ in range(args.procs)
         ]
         lats: list[float] = []
-        total = matched = 0
+        total = matched = errors = 0
         wall = 0.0
         for c in clients:
             out, _ = c.communicate(timeout=args.seconds + 600)
+            if not out.strip():
+                print("client produced no output (crashed)", file=sys.stderr)
+                errors += 1
+                continue
             d = json.loads(out.splitlines()[-1])
             total += d["count"]
             matched += d["matched"]
+            errors += d.get("errors", 0)
             wall = max(wall, d["wall"])
             lats.extend(d["lats_ms"])
         lats.sort()
@@ -152,6 +168,7 @@ def main() -> int:
             "value": round(total / wall, 1),
             "requests": total,
             "matched": matched,
+            "errors": errors,
             "p50_ms": lats[len(lats) // 2] if lats else None,
             "p99_ms": lats[max(0, int(len(lats) * 0.99) - 1)] if lats else None,
             "entries": args.entries,
