@@ -528,3 +528,20 @@ def test_smallb_v4_matches_default():
         assert r.returncode == 0, r.stderr[-1500:]
         outs[sel] = float(r.stdout.split("CSUM")[1].strip())
     assert abs(outs[None] - outs["2"]) < 1e-2, outs
+
+
+@pytest.mark.gpu
+def test_smallb_dim_tail_segments():
+    """smallb v4 streams rows in 4-segment (1 KiB) batches; D % 256 != 0
+    leaves tail segments (e.g. D=832 -> 13 segments = 3 batches + 1
+    tail). Must stay exact vs fp32 torch on that path too."""
+    from kakveda_amd import ops
+
+    q = _rand_unit(2, 832, seed=210)
+    c = _rand_unit(70000, 832, seed=211)
+    scores, idx = ops.cosine_topk(q, c, 5)
+    torch.cuda.synchronize()
+    sims = q.float() @ c.float().t()
+    ref_s, _ = torch.topk(sims, 5, dim=1)
+    assert torch.allclose(scores, ref_s, atol=2e-2, rtol=1e-2)
+    assert torch.allclose(sims.gather(1, idx), ref_s, atol=1e-4)
