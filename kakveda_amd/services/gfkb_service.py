@@ -138,6 +138,18 @@ def create_app(
     # the event loop — a slow match would stall every concurrent request
     # including /healthz (matches the reference's sync-handler behaviour)
 
+    # anyio's default 40-token thread limiter caps how many requests can
+    # be parked inside `match` waiting on the batcher, which caps the
+    # micro-batch size and hence service throughput. Raise it so the
+    # batcher, not the threadpool, sets the batch size. (The front-door
+    # warn app registers the same hook: ASGITransport composition never
+    # runs THIS app's lifespan.)
+    @app.on_event("startup")
+    async def _raise_threadpool():
+        from kakveda_amd.services.wiring import raise_thread_limiter
+
+        raise_thread_limiter()
+
     @app.get("/failures")
     def list_failures():
         return {"failures": engine.list_failures()}

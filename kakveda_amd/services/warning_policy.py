@@ -33,6 +33,15 @@ def create_app(
     cfg = config or ConfigStore()
     app.state.transport = tx
 
+    # front door: raise the sync-handler thread limiter for this event
+    # loop (covers the locally-composed gfkb app too — its own lifespan
+    # never runs under ASGITransport). See wiring.raise_thread_limiter.
+    @app.on_event("startup")
+    async def _raise_threadpool():
+        from kakveda_amd.services.wiring import raise_thread_limiter
+
+        raise_thread_limiter()
+
     @app.post("/warn", response_model=WarningResponse)
     async def warn(req: WarningRequest):
         threshold = float(cfg.get("failure_matching.similarity_threshold", 0.8))

@@ -16,6 +16,28 @@ from typing import Any, Dict, Optional
 import httpx
 
 
+def raise_thread_limiter(default: int = 256) -> int:
+    """Raise anyio's default 40-token sync-handler thread limiter for the
+    CURRENT event loop (call from an app startup hook).
+
+    FastAPI runs plain-`def` handlers through this limiter; 40 tokens cap
+    how many requests can be parked inside the gfkb `match` handler
+    waiting on the micro-batcher, which caps the batch size and hence
+    service throughput (measured: the HTTP peak sat at exactly
+    40/cycle/worker — profiles/serving_http.md). KAKVEDA_THREADPOOL
+    overrides the new cap."""
+    import os
+
+    tokens = int(os.environ.get("KAKVEDA_THREADPOOL", str(default)))
+    try:
+        import anyio.to_thread
+
+        anyio.to_thread.current_default_thread_limiter().total_tokens = tokens
+    except Exception:  # pragma: no cover - anyio internals moved
+        return 0
+    return tokens
+
+
 class Transport:
     def __init__(self, timeout: float = 3.0):
         self.timeout = timeout

@@ -211,3 +211,26 @@ def test_http_bench_harness_runs():
     assert out["requests"] > 0 and out["value"] > 0
     assert out["matched"] > 0  # the seeded failure matches the demo prompt
     assert out["p99_ms"] >= out["p50_ms"] > 0
+
+
+def test_thread_limiter_raised_on_startup():
+    """The warn app's startup hook must raise anyio's default 40-token
+    sync-handler limiter (it capped concurrent batcher occupancy and
+    hence service throughput)."""
+    import anyio.to_thread
+    from fastapi.testclient import TestClient
+
+    from kakveda_amd.services.warning_policy import create_app
+
+    app = create_app(gfkb_url="http://gfkb:8101")
+    captured = {}
+
+    @app.get("/__limiter")
+    async def limiter():  # runs on the app's event loop
+        captured["tokens"] = anyio.to_thread.current_default_thread_limiter().total_tokens
+        return {"tokens": captured["tokens"]}
+
+    with TestClient(app) as client:  # context manager runs lifespan
+        r = client.get("/__limiter")
+        assert r.status_code == 200
+    assert captured["tokens"] == 256, captured
