@@ -46,6 +46,10 @@ if _AVAILABLE:
     WARN_DECISIONS = Counter(
         "kakveda_warn_decisions_total", "Pre-flight warning decisions", ["action"]
     )
+    WARN_SHED = Counter(
+        "kakveda_warn_shed_total",
+        "Pre-flight requests shed by admission control (503)",
+    )
 
 
 def _route_template(request: Request) -> str:
@@ -84,3 +88,8 @@ def observe_gfkb(rows: int, failure_records: int) -> None:
 def observe_warn(action: Optional[str]) -> None:
     if _AVAILABLE and action:
         WARN_DECISIONS.labels(action).inc()
+
+
+def observe_shed() -> None:
+    if _AVAILABLE:
+        WARN_SHED.inc()

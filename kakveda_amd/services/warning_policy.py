@@ -12,7 +12,7 @@ from __future__ import annotations
 import os
 from typing import Optional
 
-from fastapi import FastAPI
+from fastapi import FastAPI, HTTPException
 
 from kakveda_amd.core.config import ConfigStore
 from kakveda_amd.core.schemas import WarningRequest, WarningResponse
@@ -42,8 +42,33 @@ def create_app(
 
         raise_thread_limiter()
 
+    # admission control: past ~512 outstanding requests per worker the
+    # service degrades non-linearly (measured overload knee —
+    # profiles/serving_http.md), so shed load with a fast 503 +
+    # Retry-After instead of letting every request queue into the knee.
+    # KAKVEDA_MAX_INFLIGHT=0 disables.
+    max_inflight = int(os.environ.get("KAKVEDA_MAX_INFLIGHT", "512"))
+    inflight = {"n": 0}
+    app.state.inflight = inflight
+
     @app.post("/warn", response_model=WarningResponse)
     async def warn(req: WarningRequest):
+        if max_inflight > 0 and inflight["n"] >= max_inflight:
+            from kakveda_amd.core.metrics import observe_shed
+
+            observe_shed()
+            raise HTTPException(
+                status_code=503,
+                detail="warning-policy at capacity; retry",
+                headers={"Retry-After": "1"},
+            )
+        inflight["n"] += 1
+        try:
+            return await _warn(req)
+        finally:
+            inflight["n"] -= 1
+
+    async def _warn(req: WarningRequest) -> WarningResponse:
         threshold = float(cfg.get("failure_matching.similarity_threshold", 0.8))
         action_default = str(cfg.get("warning_policy.default_action", "warn"))
 

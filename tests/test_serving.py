@@ -234,3 +234,65 @@ def test_thread_limiter_raised_on_startup():
         r = client.get("/__limiter")
         assert r.status_code == 200
     assert captured["tokens"] == 256, captured
+
+
+def test_warn_admission_control(monkeypatch):
+    """Past KAKVEDA_MAX_INFLIGHT outstanding /warn requests the front
+    door sheds load with 503 + Retry-After instead of queueing into the
+    measured overload knee (profiles/serving_http.md)."""
+    import anyio
+    from fastapi.testclient import TestClient
+
+    monkeypatch.setenv("KAKVEDA_MAX_INFLIGHT", "1")
+    from kakveda_amd.services.warning_policy import create_app as warn_app
+    from kakveda_amd.services.wiring import Transport
+
+    # a gfkb stand-in whose /failures/match blocks until released, so the
+    # first request holds the in-flight slot while the second arrives
+    import threading
+
+    from fastapi import FastAPI
+
+    gate = threading.Event()  # thread-safe across the client's event loops
+    stub = FastAPI()
+
+    @stub.post("/failures/match")
+    async def match(body: dict):
+        await anyio.to_thread.run_sync(gate.wait)
+        return {"matches": []}
+
+    @stub.get("/patterns")
+    async def patterns():
+        return {"patterns": []}
+
+    tx = Transport()
+    tx.register_local("http://gfkb:8101", stub)
+    app = warn_app(gfkb_url="http://gfkb:8101", transport=tx)
+
+    body = {"app_id": "a", "prompt": "p", "tools": [], "env": {}}
+    with TestClient(app) as client:
+        import threading
+
+        results = {}
+
+        def first():
+            results["first"] = client.post("/warn", json=body).status_code
+
+        t = threading.Thread(target=first)
+        t.start()
+        # wait until the first request occupies the slot
+        import time as _t
+
+        for _ in range(200):
+            if app.state.inflight["n"] >= 1:
+                break
+            _t.sleep(0.01)
+        assert app.state.inflight["n"] == 1
+        r2 = client.post("/warn", json=body)
+        assert r2.status_code == 503
+        assert r2.headers.get("retry-after") == "1"
+        gate.set()
+        t.join(timeout=10)
+        assert results["first"] == 200
+    # counter drained
+    assert app.state.inflight["n"] == 0
