@@ -57,16 +57,25 @@ def client_main(url: str, seconds: float, conns: int) -> None:
                 return_exceptions=True)
             stop_at = time.perf_counter() + seconds
 
+            sheds = 0
+
             async def loop() -> None:
-                nonlocal matched, errors
+                nonlocal matched, errors, sheds
                 while time.perf_counter() < stop_at:
                     t0 = time.perf_counter()
                     try:
                         r = await cli.post("/warn", json=BODY)
-                        lats.append(time.perf_counter() - t0)
-                        if r.status_code == 200 and r.json().get("references"):
-                            matched += 1
-                        elif r.status_code != 200:
+                        if r.status_code == 200:
+                            lats.append(time.perf_counter() - t0)
+                            if r.json().get("references"):
+                                matched += 1
+                        elif r.status_code == 503:
+                            # honour admission control: back off as a
+                            # well-behaved client would
+                            sheds += 1
+                            await asyncio.sleep(
+                                float(r.headers.get("retry-after", "1")))
+                        else:
                             errors += 1
                     except Exception:
                         errors += 1
@@ -81,6 +90,7 @@ def client_main(url: str, seconds: float, conns: int) -> None:
             "wall": wall,
             "matched": matched,
             "errors": errors,
+            "sheds": sheds,
             "lats_ms": [round(x * 1000, 3) for x in lats],
         }))
 
@@ -147,7 +157,7 @@ def main() -> int:
             for _ in range(args.procs)
         ]
         lats: list[float] = []
-        total = matched = errors = 0
+        total = matched = errors = sheds = 0
         wall = 0.0
         for c in clients:
             out, _ = c.communicate(timeout=args.seconds + 600)
@@ -159,6 +169,7 @@ def main() -> int:
             total += d["count"]
             matched += d["matched"]
             errors += d.get("errors", 0)
+            sheds += d.get("sheds", 0)
             wall = max(wall, d["wall"])
             lats.extend(d["lats_ms"])
         lats.sort()
@@ -169,6 +180,7 @@ def main() -> int:
             "requests": total,
             "matched": matched,
             "errors": errors,
+            "sheds": sheds,
             "p50_ms": lats[len(lats) // 2] if lats else None,
             "p99_ms": lats[max(0, int(len(lats) * 0.99) - 1)] if lats else None,
             "entries": args.entries,
