@@ -1612,18 +1612,19 @@ __global__ __launch_bounds__(256) void smallb_emit_kernel(
   }
 }
 
-// v4 lane remap (KAKVEDA_SMALLB=4): one wave covers 8 CONSECUTIVE rows
-// with 8 lanes per row, so a single load instruction touches 8 FULL
-// 128-byte lines. v2's per-lane-row mapping touches 64 lines per
-// instruction, 16 B each — a line must survive 7 more instructions in
-// L1 to be fully consumed, and with 8 resident waves the combined
-// window exceeds L1, so lines get refetched from L2 (the suspected
-// ~4.7-of-8 TB/s efficiency gap at 100M). Scores are reduced across the
-// 8 chunk lanes with 3 shfl_xor steps at row end; per-lane f32x4
-// sub-chains keep the FMA dependency depth at 2 per segment (the v1
-// lesson). Exactness: identical emission contract to v2 — every score
-// >= the shared prepass floor is emitted; tail rows are clamped for the
-// load and guarded at emission.
+// v4 lane remap (the serving DEFAULT; KAKVEDA_SMALLB=2 reverts): one
+// wave covers 8 CONSECUTIVE rows with 8 lanes per row, so a single load
+// instruction touches 8 FULL 128-byte lines, and scores are reduced
+// across the 8 chunk lanes with 3 shfl_xor steps at row end; per-lane
+// f32x4 sub-chains keep the FMA dependency depth at 2 per segment (the
+// v1 lesson). Measured +13% over v2 at B=4 (parity at B=1), and TCC
+// counters show WHY (profiles/pmc_round2.md): L2 traffic is identical
+// (one request per 128-B line in both mappings — L1 coalesces v2's
+// eight same-line 16-B loads), so the win is the leaner inner loop
+// (1,587 vs 2,442 instructions) and 134-vs-186 VGPRs = 3-vs-2
+// waves/SIMD occupancy, NOT cache behaviour. Exactness: identical
+// emission contract to v2 — every score >= the shared prepass floor is
+// emitted; tail rows are clamped for the load and guarded at emission.
 __global__ __launch_bounds__(256) void smallb_emit_kernel_v4(
     const bf16_t* __restrict__ Q, const bf16_t* __restrict__ C, int B,
     long N, int D, const unsigned* __restrict__ rowthr,
