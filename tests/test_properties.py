@@ -1,0 +1,186 @@
+"""Property-based tests (hypothesis) for the exact-semantics components:
+signature canonicalisation, feature hashing, the incremental health
+scorer vs a brute-force recompute, and segmented-store search parity.
+
+These guard the invariants the services rely on but example-based tests
+only spot-check: canonical-form idempotence, order-insensitivity of the
+signature inputs, O(1)-window bookkeeping == from-scratch recompute, and
+segmentation-invariance of search results.
+"""
+
+from __future__ import annotations
+
+import string
+from collections import Counter, deque
+
+import numpy as np
+import pytest
+import torch
+from hypothesis import given, settings, strategies as st
+
+# text strategies kept ASCII-printable: the wire contract normalises
+# case/whitespace, and non-ASCII case-folding is out of contract
+_words = st.text(alphabet=string.ascii_letters + string.digits + ".,;:!?", min_size=1, max_size=12)
+_prompts = st.lists(_words, min_size=0, max_size=20).map(" ".join)
+_ws = st.sampled_from([" ", "  ", "\t", "\n", " \t "])
+
+
+# --------------------------------------------------------------------------
+# signature canonicalisation
+# --------------------------------------------------------------------------
+
+
+@settings(max_examples=60, deadline=None)
+@given(_prompts, _ws, _ws)
+def test_normalize_whitespace_and_case_invariant(p, lead, sep):
+    from kakveda_amd.core.signature import normalize_prompt
+
+    messy = lead + p.upper().replace(" ", sep) + lead
+    assert normalize_prompt(messy) == normalize_prompt(p)
+    # idempotence
+    assert normalize_prompt(normalize_prompt(p)) == normalize_prompt(p)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    _prompts,
+    st.lists(_words, max_size=6),
+    st.dictionaries(_words, st.one_of(st.integers(), _words), max_size=6),
+)
+def test_signature_text_order_insensitive(p, tools, env):
+    from kakveda_amd.core.signature import fingerprint, signature_text
+
+    s = signature_text(p, tools, env)
+    assert s.startswith("intent_tags:")
+    for section in (" | prompt_hint:", " | tools:", " | env_keys:"):
+        assert section in s
+    # tools: order and duplicates are canonicalised away
+    assert signature_text(p, list(reversed(tools)) + tools, env) == s
+    # env: only keys matter, never values
+    env2 = {k: "DIFFERENT" for k in env}
+    assert signature_text(p, tools, env2) == s
+    fp = fingerprint(p, tools, env)
+    assert len(fp) == 16 and all(c in "0123456789abcdef" for c in fp)
+    assert fingerprint(p, list(reversed(tools)), env2) == fp
+
+
+@settings(max_examples=40, deadline=None)
+@given(_prompts)
+def test_intent_tags_sorted_and_stable(p):
+    from kakveda_amd.core.signature import prompt_intent_tags
+
+    tags = prompt_intent_tags(p)
+    assert tags == sorted(tags)
+    assert prompt_intent_tags(p) == tags
+    # tags survive arbitrary re-casing (contract: case-insensitive)
+    assert prompt_intent_tags(p.upper()) == tags
+
+
+# --------------------------------------------------------------------------
+# feature hashing
+# --------------------------------------------------------------------------
+
+
+@settings(max_examples=50, deadline=None)
+@given(_prompts, st.sampled_from([256, 4096, 1 << 16]))
+def test_featurize_invariants(text, hash_dim):
+    from kakveda_amd.encoder.featurizer import featurize
+
+    idxs, ws = featurize(text, hash_dim=hash_dim)
+    assert idxs.dtype == np.int32 and ws.dtype == np.float32
+    assert len(idxs) == len(ws)
+    if len(idxs):
+        assert (np.diff(idxs) > 0).all()  # sorted unique
+        assert idxs.min() >= 0 and idxs.max() < hash_dim
+        assert abs(float(np.linalg.norm(ws)) - 1.0) < 1e-5
+    # determinism
+    i2, w2 = featurize(text, hash_dim=hash_dim)
+    assert np.array_equal(idxs, i2) and np.array_equal(ws, w2)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.lists(_prompts, min_size=1, max_size=5), st.sampled_from([4, 16, 64]))
+def test_featurize_batch_shape_and_norm(texts, max_features):
+    from kakveda_amd.encoder.featurizer import featurize_batch
+
+    idx, w = featurize_batch(texts, max_features=max_features)
+    assert idx.shape == (len(texts), max_features)
+    assert w.shape == (len(texts), max_features)
+    for b in range(len(texts)):
+        n = float(np.linalg.norm(w[b]))
+        assert n == pytest.approx(1.0, abs=1e-5) or n == 0.0
+
+
+# --------------------------------------------------------------------------
+# incremental health scorer == brute-force recompute
+# --------------------------------------------------------------------------
+
+_events = st.lists(
+    st.fixed_dictionaries(
+        {
+            "app_id": st.sampled_from(["a", "b", "c"]),
+            "severity": st.sampled_from(["low", "medium", "high", "unknown"]),
+            "failure_type": st.sampled_from(["T1", "T2", "T3", "T4"]),
+        }
+    ),
+    min_size=1,
+    max_size=80,
+)
+
+
+@settings(max_examples=40, deadline=None)
+@given(_events, st.sampled_from([1, 3, 7, 50]))
+def test_health_scorer_matches_bruteforce(events, window):
+    from kakveda_amd.health.scoring import _DEFAULT_WEIGHTS, HealthScorer
+
+    scorer = HealthScorer(window_size=window)
+    per_app: dict[str, deque] = {}
+    for ev in events:
+        point = scorer.observe(dict(ev))
+        win = per_app.setdefault(ev["app_id"], deque(maxlen=window))
+        win.append(ev)
+        # brute-force recompute of the window aggregates
+        weighted = sum(_DEFAULT_WEIGHTS.get(e["severity"], 1.0) for e in win)
+        counts = Counter(e["failure_type"] for e in win)
+        penalty = 2.5 * sum(max(0, c - 1) for c in counts.values())
+        expected = max(0.0, 100.0 - weighted * 5.0 - penalty)
+        assert point.score == pytest.approx(expected, abs=1e-9)
+        assert point.recurrent_penalty == pytest.approx(penalty, abs=1e-9)
+        assert point.notes["window_failures"] == len(win)
+        assert point.notes["weighted"] == pytest.approx(weighted, abs=1e-9)
+        assert point.notes["last_failure"] == ev["failure_type"]
+
+
+# --------------------------------------------------------------------------
+# segmented store: search is segmentation-invariant (CPU fallback path)
+# --------------------------------------------------------------------------
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    st.lists(st.integers(min_value=1, max_value=40), min_size=1, max_size=8),
+    st.sampled_from([16, 33]),
+)
+def test_segmented_store_search_matches_flat(chunks, segment_rows):
+    from kakveda_amd.gfkb.engine import EmbeddingStore
+
+    dim = 32
+    g = torch.Generator().manual_seed(sum(chunks) * 31 + segment_rows)
+    rows = torch.randn(sum(chunks), dim, generator=g)
+    rows = rows / rows.norm(dim=-1, keepdim=True)
+
+    store = EmbeddingStore(dim, device="cpu", capacity=8, segment_rows=segment_rows)
+    s = 0
+    for c in chunks:
+        store.append(rows[s : s + c])
+        s += c
+    assert store.count == sum(chunks)
+
+    q = torch.randn(3, dim, generator=g)
+    q = q / q.norm(dim=-1, keepdim=True)
+    k = min(5, sum(chunks))
+    scores, idx = store.search(q, k)
+    sims = q @ rows.t()
+    ref_s, _ = torch.topk(sims, k, dim=1)
+    assert torch.allclose(scores, ref_s, atol=1e-5), (scores - ref_s).abs().max()
+    assert torch.allclose(sims.gather(1, idx), ref_s, atol=1e-5)
