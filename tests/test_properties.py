@@ -184,3 +184,68 @@ def test_segmented_store_search_matches_flat(chunks, segment_rows):
     ref_s, _ = torch.topk(sims, k, dim=1)
     assert torch.allclose(scores, ref_s, atol=1e-5), (scores - ref_s).abs().max()
     assert torch.allclose(sims.gather(1, idx), ref_s, atol=1e-5)
+
+
+# --------------------------------------------------------------------------
+# engine versioned upsert + restart == replay (append-only log invariants)
+# --------------------------------------------------------------------------
+
+_upserts = st.lists(
+    st.tuples(
+        st.sampled_from(["T1", "T2"]),
+        st.sampled_from(["alpha beta", "gamma delta", "epsilon zeta"]),
+        st.sampled_from(["app1", "app2"]),
+    ),
+    min_size=1,
+    max_size=12,
+)
+
+
+@settings(max_examples=10, deadline=None)
+@given(_upserts)
+def test_engine_versioning_and_restart_replay(seq):
+    import tempfile
+
+    from kakveda_amd.gfkb.engine import GfkbEngine
+
+    with tempfile.TemporaryDirectory() as td:
+        eng = GfkbEngine(data_dir=td, device="cpu", dim=32, hash_dim=256)
+        mirror: dict = {}
+        for ft, sig, app in seq:
+            rec, created = eng.upsert_failure(ft, sig, {"m": 1}, app_id=app)
+            key = (ft, sig)
+            prev = mirror.get(key)
+            assert created == (prev is None)
+            if prev is None:
+                mirror[key] = {"id": rec["failure_id"], "version": 1,
+                               "occ": 1, "apps": [app]}
+            else:
+                prev["version"] += 1
+                prev["occ"] += 1
+                if app not in prev["apps"]:
+                    prev["apps"].append(app)
+            m = mirror[key]
+            assert rec["failure_id"] == m["id"]  # identity keeps its id
+            assert rec["version"] == m["version"]
+            assert rec["occurrences"] == m["occ"]
+            assert rec["affected_apps"] == m["apps"]
+        ids = {m["id"] for m in mirror.values()}
+        assert len(ids) == len(mirror)  # distinct identities, distinct ids
+
+        # the append-only log has one record per upsert
+        assert len(eng.failures) == len(seq)
+
+        # restart: a fresh engine over the same data_dir replays to the
+        # same latest records and one store row per identity
+        eng2 = GfkbEngine(data_dir=td, device="cpu", dim=32, hash_dim=256)
+        assert eng2.store.count == len(mirror)
+        for (ft, sig), m in mirror.items():
+            rec2 = eng2._latest[(ft, sig)]
+            assert rec2["failure_id"] == m["id"]
+            assert rec2["version"] == m["version"]
+            assert rec2["occurrences"] == m["occ"]
+            # self-match ranks the identity first with ~perfect score
+            matches = eng2.match(sig, failure_type=ft)
+            assert matches and matches[0].failure_id == m["id"]
+            assert matches[0].score > 0.99
+            assert matches[0].version == m["version"]
